@@ -1,0 +1,387 @@
+"""Public API: init / remote / get / kill / shutdown.
+
+Parity: /root/reference/fed/api.py (init :67-296, shutdown :299-360,
+FedRemoteFunction :384-416, FedRemoteClass :419-448, remote :452-528,
+get :531-608, kill :611-623).  Same multi-controller contract: every party
+runs the identical driver script; this module decides per call what runs
+locally and what becomes a cross-party push/recv pair.
+"""
+from __future__ import annotations
+
+import functools
+import inspect
+import logging
+import signal
+import sys
+import threading
+from typing import Any, Callable, Dict, List, Optional, Union
+
+import cloudpickle
+
+from rayfed_amd import config as fed_config
+from rayfed_amd._private import constants, kv as kv_mod
+from rayfed_amd._private.call_holder import FedCallHolder
+from rayfed_amd._private.fed_actor import FedActorHandle, _invoke_materialized
+from rayfed_amd._private.global_context import (
+    clear_global_context,
+    get_global_context,
+    init_global_context,
+)
+from rayfed_amd.exceptions import FedRemoteError
+from rayfed_amd.fed_object import FedObject
+from rayfed_amd.proxy import barriers
+from rayfed_amd.runtime.object_ref import ObjectRef
+from rayfed_amd.utils import is_cython, setup_logger, validate_addresses
+
+logger = logging.getLogger(__name__)
+
+_original_sigint_handler = None
+
+
+def _signal_handler(signum, frame):
+    if signum == signal.SIGINT:
+        logger.warning("Stop the fed on SIGINT (failure-triggered shutdown).")
+        _shutdown(intended=False)
+
+
+def init(
+    addresses: Dict = None,
+    party: str = None,
+    config: Dict = None,
+    tls_config: Dict = None,
+    logging_level: str = "info",
+    sender_proxy_cls=None,
+    receiver_proxy_cls=None,
+    receiver_sender_proxy_cls=None,
+    job_name: str = None,
+    sending_failure_handler: Callable[[Exception], None] = None,
+):
+    """Initialize the fed runtime for one party.
+
+    Mirrors the reference's ``fed.init`` signature and observable behavior
+    (api.py:67-296); the substrate underneath is the in-process MI355X
+    runtime instead of Ray.
+    """
+    assert addresses, "addresses must be provided"
+    assert party, "party must be provided"
+    assert party in addresses, f"party {party!r} is not in the addresses"
+    validate_addresses(addresses)
+    config = config or {}
+
+    if job_name is None:
+        job_name = constants.RAYFED_DEFAULT_JOB_NAME
+
+    cross_silo_comm_dict = config.get("cross_silo_comm", {})
+    cross_silo_comm_config = fed_config.GrpcCrossSiloMessageConfig.from_dict(
+        cross_silo_comm_dict
+    )
+    gpu_plane_dict = config.get("gpu_data_plane", {})
+
+    ctx = init_global_context(
+        current_party=party,
+        job_name=job_name,
+        sending_failure_handler=sending_failure_handler,
+        exit_on_sending_failure=cross_silo_comm_config.exit_on_sending_failure,
+        continue_waiting_for_data_sending_on_error=(
+            cross_silo_comm_config.continue_waiting_for_data_sending_on_error
+        ),
+    )
+
+    # Persist cluster/job config in the internal KV — parity with
+    # api.py:204-218 (cloudpickled dicts under the job-prefixed keys).
+    kv = kv_mod._init_internal_kv(job_name)
+    cluster_config = {
+        constants.KEY_OF_CLUSTER_ADDRESSES: addresses,
+        constants.KEY_OF_CURRENT_PARTY_NAME: party,
+        constants.KEY_OF_TLS_CONFIG: tls_config,
+    }
+    job_config = {
+        constants.KEY_OF_CROSS_SILO_COMM_CONFIG_DICT: cross_silo_comm_dict,
+    }
+    kv.put(constants.KEY_OF_CLUSTER_CONFIG, cloudpickle.dumps(cluster_config))
+    kv.put(constants.KEY_OF_JOB_CONFIG, cloudpickle.dumps(job_config))
+    fed_config._clear_cached_config()
+
+    setup_logger(logging_level=logging_level, party=party, job_name=job_name)
+    logger.info("Started rayfed-amd with %s", cluster_config)
+
+    global _original_sigint_handler
+    if threading.current_thread() is threading.main_thread():
+        _original_sigint_handler = signal.signal(signal.SIGINT, _signal_handler)
+
+    ctx.get_cleanup_manager().start(
+        exit_on_sending_failure=cross_silo_comm_config.exit_on_sending_failure,
+        expose_error_trace=cross_silo_comm_config.expose_error_trace,
+        continue_waiting_for_data_sending_on_error=(
+            cross_silo_comm_config.continue_waiting_for_data_sending_on_error
+        ),
+    )
+
+    use_global_proxy = (
+        cross_silo_comm_config.use_global_proxy
+        if cross_silo_comm_config.use_global_proxy is not None
+        else True
+    )
+    receiver = barriers.start_receiver_proxy(
+        addresses,
+        party,
+        job_name=job_name,
+        tls_config=tls_config,
+        proxy_cls=receiver_proxy_cls,
+        proxy_config=cross_silo_comm_config,
+        use_global_proxy=use_global_proxy,
+    )
+    sender = barriers.start_sender_proxy(
+        addresses,
+        party,
+        job_name=job_name,
+        tls_config=tls_config,
+        proxy_cls=sender_proxy_cls,
+        proxy_config=cross_silo_comm_config,
+        use_global_proxy=use_global_proxy,
+    )
+
+    # Attach the GPU data plane when a HIP device is visible: tensors then
+    # ride the pack/CRC/pinned-staging path instead of pickle.
+    try:
+        from rayfed_amd.ops.gpu_plane import maybe_create_gpu_plane
+
+        plane = maybe_create_gpu_plane(gpu_plane_dict)
+        if plane is not None:
+            sender.proxy.gpu_plane = plane
+            receiver.proxy.gpu_plane = plane
+    except Exception:  # noqa: BLE001 - the control plane must work CPU-only
+        logger.debug("GPU data plane unavailable", exc_info=True)
+
+    if config.get("barrier_on_initializing", False):
+        barriers.ping_others(addresses=addresses, self_party=party)
+
+
+def shutdown():
+    """Intended shutdown: flush pending cross-party sends, then tear down."""
+    _shutdown(intended=True)
+
+
+def _shutdown(intended: bool = True):
+    """Parity: api.py:299-360 — on unintended exit run the failure handler,
+    optionally abandon pending sends, and exit(1)."""
+    ctx = get_global_context()
+    if ctx is None:
+        return
+
+    if intended:
+        wait_for_sending = True
+    else:
+        wait_for_sending = ctx.get_continue_waiting_for_data_sending_on_error()
+        handler = ctx.get_sending_failure_handler()
+        if handler is not None:
+            try:
+                handler(ctx.get_last_received_error())
+            except Exception:  # noqa: BLE001
+                logger.exception("sending_failure_handler raised")
+
+    clear_global_context(wait_for_sending=wait_for_sending)
+    barriers._cleanup_proxies()
+    kv_mod._clear_internal_kv()
+    fed_config._clear_cached_config()
+
+    global _original_sigint_handler
+    if (
+        _original_sigint_handler is not None
+        and threading.current_thread() is threading.main_thread()
+    ):
+        signal.signal(signal.SIGINT, _original_sigint_handler)
+        _original_sigint_handler = None
+
+    logger.info("Shutdown rayfed-amd (intended=%s).", intended)
+    if not intended:
+        sys.exit(1)
+
+
+def _get_addresses(job_name: str = None):
+    cc = fed_config.get_cluster_config()
+    return cc.cluster_addresses if cc else None
+
+
+def _get_party(job_name: str = None):
+    cc = fed_config.get_cluster_config()
+    return cc.current_party if cc else None
+
+
+def _get_tls(job_name: str = None):
+    cc = fed_config.get_cluster_config()
+    return cc.tls_config if cc else None
+
+
+class FedRemoteFunction:
+    def __init__(self, func_or_class) -> None:
+        self._node_party = None
+        self._func_body = func_or_class
+        self._options = {}
+        self._fed_call_holder = None
+
+    def party(self, party: str) -> "FedRemoteFunction":
+        self._node_party = party
+        self._fed_call_holder = FedCallHolder(
+            party, self._execute_impl, self._options
+        )
+        return self
+
+    def options(self, **options) -> "FedRemoteFunction":
+        self._options = options
+        if self._fed_call_holder:
+            self._fed_call_holder.options(**options)
+        return self
+
+    def remote(self, *args, **kwargs):
+        if not self._node_party:
+            raise ValueError(
+                "You should specify a party by using `.party(...)` before "
+                "`.remote()`."
+            )
+        return self._fed_call_holder.internal_remote(*args, **kwargs)
+
+    def _execute_impl(self, args, kwargs):
+        ctx = get_global_context()
+        return ctx.get_executor().submit(
+            _invoke_materialized,
+            args=(self._func_body, args, kwargs),
+            num_returns=self._options.get("num_returns", 1),
+        )
+
+
+class FedRemoteClass:
+    def __init__(self, func_or_class) -> None:
+        self._party = None
+        self._cls = func_or_class
+        self._options = {}
+
+    def party(self, party: str) -> "FedRemoteClass":
+        self._party = party
+        return self
+
+    def options(self, **options) -> "FedRemoteClass":
+        self._options = options
+        return self
+
+    def remote(self, *cls_args, **cls_kwargs) -> FedActorHandle:
+        if not self._party:
+            raise ValueError(
+                "You should specify a party by using `.party(...)` before "
+                "`.remote()`."
+            )
+        ctx = get_global_context()
+        fed_class_task_id = ctx.next_seq_id()
+        handle = FedActorHandle(
+            fed_class_task_id,
+            _get_addresses(),
+            self._cls,
+            ctx.get_current_party(),
+            self._party,
+            self._options,
+            ctx.get_executor(),
+        )
+        call_holder = FedCallHolder(self._party, handle._execute_impl, self._options)
+        call_holder.internal_remote(*cls_args, **cls_kwargs)
+        return handle
+
+
+def remote(*args, **kwargs):
+    """``@fed.remote`` decorator for functions and classes
+    (parity: api.py:452-528)."""
+
+    def _make_fed_remote(func_or_class, **options):
+        if inspect.isfunction(func_or_class) or is_cython(func_or_class):
+            fr = FedRemoteFunction(func_or_class)
+            if options:
+                fr.options(**options)
+            return fr
+        if inspect.isclass(func_or_class):
+            fc = FedRemoteClass(func_or_class)
+            if options:
+                fc.options(**options)
+            return fc
+        raise TypeError(
+            "The @fed.remote decorator must be applied to a function or a class."
+        )
+
+    if len(args) == 1 and len(kwargs) == 0 and callable(args[0]):
+        return _make_fed_remote(args[0])
+    assert len(args) == 0 and len(kwargs) > 0, "Remote args error."
+    return functools.partial(_make_fed_remote, **kwargs)
+
+
+def get(
+    fed_objects: Union[FedObject, ObjectRef, List],
+) -> Any:
+    """Fetch values; on owned objects *broadcast* them to every other party —
+    the symmetric-execution rule that keeps all drivers' DAGs aligned
+    (parity: api.py:531-608)."""
+    if isinstance(fed_objects, ObjectRef):
+        return fed_objects.result()
+
+    is_individual_id = isinstance(fed_objects, FedObject)
+    if is_individual_id:
+        fed_objects = [fed_objects]
+
+    ctx = get_global_context()
+    if ctx is None:
+        raise RuntimeError("fed.init must be called before fed.get")
+    addresses = _get_addresses()
+    current_party = ctx.get_current_party()
+    fake_fed_task_id = ctx.next_seq_id()
+
+    refs: List[ObjectRef] = []
+    for fed_object in fed_objects:
+        if not isinstance(fed_object, FedObject):
+            if isinstance(fed_object, ObjectRef):
+                refs.append(fed_object)
+                continue
+            raise TypeError(
+                f"fed.get expects FedObjects, got {type(fed_object).__name__}"
+            )
+        if fed_object.get_party() == current_party:
+            ref = fed_object.get_ray_object_ref()
+            assert ref is not None
+            refs.append(ref)
+            for party_name in addresses:
+                if party_name == current_party:
+                    continue
+                if fed_object.was_sending_or_sent_to_party(party_name):
+                    continue
+                fed_object.mark_is_sending_to_party(party_name)
+                barriers.send(
+                    dest_party=party_name,
+                    data=ref,
+                    upstream_seq_id=fed_object.get_fed_task_id(),
+                    downstream_seq_id=fake_fed_task_id,
+                )
+        else:
+            if fed_object.get_ray_object_ref() is None:
+                ref = barriers.recv(
+                    current_party,
+                    fed_object.get_party(),
+                    fed_object.get_fed_task_id(),
+                    fake_fed_task_id,
+                )
+                fed_object._cache_ray_object_ref(ref)
+            refs.append(fed_object.get_ray_object_ref())
+
+    try:
+        values = [ref.result() for ref in refs]
+    except FedRemoteError as e:
+        logger.warning("Receiving exception from a remote party: %r", e)
+        ctx.set_last_received_error(e)
+        raise
+
+    return values[0] if is_individual_id else values
+
+
+def kill(actor: FedActorHandle, *, no_restart: bool = True):
+    """Kill an actor — only effective in its owning party
+    (parity: api.py:611-623)."""
+    ctx = get_global_context()
+    if ctx is None:
+        return
+    if actor._node_party == ctx.get_current_party():
+        actor._kill()

@@ -1,0 +1,148 @@
+"""Config system: cluster/job config + cross-silo message config dataclasses.
+
+Parity: /root/reference/fed/config.py:15-195.  Same layering — ``fed.init``
+writes cloudpickled config dicts into the internal KV under the job's prefix;
+readers lazily load and cache them; ``CrossSiloMessageConfig.from_dict``
+filters unknown keys so configs round-trip across versions.
+
+Additions for the MI355X build: ``GpuDataPlaneConfig`` — chunk sizing for the
+HIP pack → pinned-staging → gRPC pipeline and CRC verification toggles.
+"""
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field, fields
+from typing import Dict, List, Optional
+
+import cloudpickle
+
+import rayfed_amd._private.constants as constants
+from rayfed_amd._private import kv as kv_mod
+
+
+class ClusterConfig:
+    """Addresses / current party / TLS, read back from the internal KV."""
+
+    def __init__(self, raw_bytes: bytes):
+        self._data = cloudpickle.loads(raw_bytes)
+
+    @property
+    def cluster_addresses(self):
+        return self._data[constants.KEY_OF_CLUSTER_ADDRESSES]
+
+    @property
+    def current_party(self):
+        return self._data[constants.KEY_OF_CURRENT_PARTY_NAME]
+
+    @property
+    def tls_config(self):
+        return self._data[constants.KEY_OF_TLS_CONFIG]
+
+
+class JobConfig:
+    def __init__(self, raw_bytes: Optional[bytes]):
+        self._data = {} if raw_bytes is None else cloudpickle.loads(raw_bytes)
+
+    @property
+    def cross_silo_comm_config_dict(self) -> Dict:
+        return self._data.get(constants.KEY_OF_CROSS_SILO_COMM_CONFIG_DICT, {})
+
+
+_cluster_config: Optional[ClusterConfig] = None
+_job_config: Optional[JobConfig] = None
+
+
+def get_cluster_config() -> Optional[ClusterConfig]:
+    """Lazily load the cluster config from the internal KV (cached)."""
+    global _cluster_config
+    if _cluster_config is None:
+        if kv_mod.kv is None:
+            return None
+        raw = kv_mod.kv.get(constants.KEY_OF_CLUSTER_CONFIG)
+        if raw is None:
+            return None
+        _cluster_config = ClusterConfig(raw)
+    return _cluster_config
+
+
+def get_job_config() -> JobConfig:
+    global _job_config
+    if _job_config is None:
+        raw = kv_mod.kv.get(constants.KEY_OF_JOB_CONFIG) if kv_mod.kv else None
+        _job_config = JobConfig(raw)
+    return _job_config
+
+
+def _clear_cached_config() -> None:
+    global _cluster_config, _job_config
+    _cluster_config = None
+    _job_config = None
+
+
+@dataclass
+class CrossSiloMessageConfig:
+    """Transport-independent cross-silo messaging knobs.
+
+    Parity: config.py:78-161 in the reference (same field set and defaults).
+    """
+
+    proxy_max_restarts: Optional[int] = None
+    timeout_in_ms: int = 60000
+    messages_max_size_in_bytes: Optional[int] = None
+    exit_on_sending_failure: Optional[bool] = False
+    continue_waiting_for_data_sending_on_error: Optional[bool] = False
+    serializing_allowed_list: Optional[Dict[str, str]] = None
+    send_resource_label: Optional[Dict[str, str]] = None
+    recv_resource_label: Optional[Dict[str, str]] = None
+    http_header: Optional[Dict[str, str]] = None
+    max_concurrency: Optional[int] = None
+    expose_error_trace: Optional[bool] = False
+    use_global_proxy: Optional[bool] = True
+
+    def __json__(self) -> str:
+        return json.dumps(self.__dict__)
+
+    @classmethod
+    def from_json(cls, json_str: str) -> "CrossSiloMessageConfig":
+        return cls(**json.loads(json_str))
+
+    @classmethod
+    def from_dict(cls, data: Optional[Dict]) -> "CrossSiloMessageConfig":
+        """Build from a dict, silently dropping unknown keys
+        (reference config.py:146-161)."""
+        data = data or {}
+        known = {f.name for f in fields(cls)}
+        return cls(**{k: v for k, v in data.items() if k in known})
+
+
+@dataclass
+class GrpcCrossSiloMessageConfig(CrossSiloMessageConfig):
+    """gRPC-specific additions (reference config.py:164-195)."""
+
+    grpc_channel_options: Optional[List] = None
+    grpc_retry_policy: Optional[Dict[str, str]] = None
+
+
+@dataclass
+class GpuDataPlaneConfig:
+    """MI355X data-plane knobs (new; no reference counterpart).
+
+    Controls the HIP pack kernel → pinned staging → gRPC streaming pipeline
+    used for torch.Tensor payloads.
+    """
+
+    # Per-chunk staging size for D2H overlap.  64 MiB keeps ≥4 chunks in
+    # flight for a 256 MiB+ tensor while staying far under gRPC's message cap.
+    chunk_bytes: int = 64 << 20
+    # Number of pinned staging buffers per direction (double/triple buffer).
+    staging_buffers: int = 4
+    # Compute + verify CRC32 of every chunk on the GPU.
+    verify_crc: bool = True
+    # Receive tensors straight back onto the GPU of the consuming party.
+    place_on_gpu: bool = True
+
+    @classmethod
+    def from_dict(cls, data: Optional[Dict]) -> "GpuDataPlaneConfig":
+        data = data or {}
+        known = {f.name for f in fields(cls)}
+        return cls(**{k: v for k, v in data.items() if k in known})

@@ -1,0 +1,332 @@
+"""Send/recv barriers and the proxy services hosting the cross-silo proxies.
+
+Parity: /root/reference/fed/proxy/barriers.py (SenderProxyActor :113-183,
+ReceiverProxyActor :186-240, starters :248-330, module send/recv :462-494,
+ping_others :497-523, proxy naming :31-85).
+
+Redesign (the control-path half of the MI355X rewrite): the
+reference hosts each proxy in a **separate Ray actor process**, so every
+cross-party send costs a driver→actor RPC plus a 0.1 s-polled ack queue.
+Here the proxies are **async services on one driver-owned asyncio I/O thread**
+— a send is a queue push onto the loop (microseconds), acks are
+future-callbacks, and the receive barrier is a future resolved directly by
+the gRPC handler.  The observable semantics (seq-id rendezvous, named
+proxies, per-job names with ``use_global_proxy=False``, op-count stats,
+error re-raise on recv) are preserved.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import threading
+import time
+from concurrent.futures import Future
+from typing import Any, Dict, Optional
+
+from rayfed_amd._private import constants
+from rayfed_amd._private.global_context import get_global_context
+from rayfed_amd.config import CrossSiloMessageConfig
+from rayfed_amd.proxy.base_proxy import ReceiverProxy, SenderProxy, SenderReceiverProxy
+from rayfed_amd.runtime.object_ref import ObjectRef
+
+logger = logging.getLogger(__name__)
+
+
+# -----------------------------------------------------------------------------
+# I/O loop thread
+# -----------------------------------------------------------------------------
+class IoLoop:
+    """A dedicated thread running the asyncio loop that hosts every proxy."""
+
+    def __init__(self, name: str = "rayfed-io"):
+        self._loop = asyncio.new_event_loop()
+        self._thread = threading.Thread(
+            target=self._run, name=name, daemon=True
+        )
+        self._started = threading.Event()
+        self._thread.start()
+        self._started.wait()
+
+    def _run(self):
+        asyncio.set_event_loop(self._loop)
+        self._loop.call_soon(self._started.set)
+        self._loop.run_forever()
+
+    @property
+    def loop(self) -> asyncio.AbstractEventLoop:
+        return self._loop
+
+    def run_coro(self, coro) -> Future:
+        return asyncio.run_coroutine_threadsafe(coro, self._loop)
+
+    def stop(self):
+        if self._loop.is_closed():
+            return
+
+        def _shutdown():
+            for task in asyncio.all_tasks(self._loop):
+                task.cancel()
+            self._loop.stop()
+
+        self._loop.call_soon_threadsafe(_shutdown)
+        self._thread.join(timeout=10)
+        if not self._loop.is_running():
+            self._loop.close()
+
+
+# -----------------------------------------------------------------------------
+# Proxy naming (parity: barriers.py:31-85)
+# -----------------------------------------------------------------------------
+_SENDER_PROXY_NAME = constants.RAYFED_DEFAULT_SENDER_PROXY_NAME
+_RECEIVER_PROXY_NAME = constants.RAYFED_DEFAULT_RECEIVER_PROXY_NAME
+_SENDER_RECEIVER_PROXY_NAME = constants.RAYFED_DEFAULT_SENDER_RECEIVER_PROXY_NAME
+
+
+def set_proxy_names(
+    sender_proxy_name: Optional[str] = None,
+    receiver_proxy_name: Optional[str] = None,
+    sender_receiver_proxy_name: Optional[str] = None,
+):
+    global _SENDER_PROXY_NAME, _RECEIVER_PROXY_NAME, _SENDER_RECEIVER_PROXY_NAME
+    if sender_proxy_name:
+        _SENDER_PROXY_NAME = sender_proxy_name
+    if receiver_proxy_name:
+        _RECEIVER_PROXY_NAME = receiver_proxy_name
+    if sender_receiver_proxy_name:
+        _SENDER_RECEIVER_PROXY_NAME = sender_receiver_proxy_name
+
+
+def sender_proxy_name(job_name: Optional[str] = None, use_global_proxy: bool = True):
+    return _SENDER_PROXY_NAME if use_global_proxy else f"{_SENDER_PROXY_NAME}-{job_name}"
+
+
+def receiver_proxy_name(job_name: Optional[str] = None, use_global_proxy: bool = True):
+    return (
+        _RECEIVER_PROXY_NAME
+        if use_global_proxy
+        else f"{_RECEIVER_PROXY_NAME}-{job_name}"
+    )
+
+
+# Named-service registry — the moral equivalent of Ray named actors.
+_service_registry: Dict[str, Any] = {}
+
+
+def get_service(name: str):
+    return _service_registry.get(name)
+
+
+# -----------------------------------------------------------------------------
+# Proxy services
+# -----------------------------------------------------------------------------
+class SenderProxyService:
+    """Hosts a SenderProxy on the I/O loop; thread-safe send entry point."""
+
+    def __init__(self, proxy: SenderProxy, io: IoLoop):
+        self._proxy = proxy
+        self._io = io
+        self._stats_lock = threading.Lock()
+        self.send_op_count = 0
+
+    async def _send_coro(self, dest_party, data, upstream_seq_id, downstream_seq_id):
+        if isinstance(data, ObjectRef):
+            # Await the producing task; a task error propagates to the send
+            # future, where the cleanup manager converts it into a
+            # FedRemoteError for the peer (reference barriers.py:147-174
+            # gets the same effect from Ray arg resolution).
+            data = await asyncio.wrap_future(data.future)
+        return await self._proxy.send(
+            dest_party, data, upstream_seq_id, downstream_seq_id
+        )
+
+    def send(self, dest_party, data, upstream_seq_id, downstream_seq_id) -> Future:
+        with self._stats_lock:
+            self.send_op_count += 1
+        return self._io.run_coro(
+            self._send_coro(dest_party, data, upstream_seq_id, downstream_seq_id)
+        )
+
+    def _get_stats(self) -> Dict[str, int]:
+        return {"send_op_count": self.send_op_count}
+
+    @property
+    def proxy(self) -> SenderProxy:
+        return self._proxy
+
+    def stop(self):
+        try:
+            self._io.run_coro(self._proxy.stop()).result(timeout=10)
+        except Exception:  # noqa: BLE001
+            logger.debug("sender proxy stop failed", exc_info=True)
+
+
+class ReceiverProxyService:
+    """Hosts a ReceiverProxy on the I/O loop; recv barrier = future."""
+
+    def __init__(self, proxy: ReceiverProxy, io: IoLoop):
+        self._proxy = proxy
+        self._io = io
+        self._stats_lock = threading.Lock()
+        self.receive_op_count = 0
+
+    def start(self, ready_timeout_second: int = 60):
+        self._io.run_coro(self._proxy.start()).result(timeout=ready_timeout_second)
+
+    def get_data(self, src_party, upstream_seq_id, curr_seq_id) -> Future:
+        with self._stats_lock:
+            self.receive_op_count += 1
+        return self._io.run_coro(
+            self._proxy.get_data(src_party, upstream_seq_id, curr_seq_id)
+        )
+
+    def _get_stats(self) -> Dict[str, int]:
+        return {"receive_op_count": self.receive_op_count}
+
+    @property
+    def proxy(self) -> ReceiverProxy:
+        return self._proxy
+
+    def stop(self):
+        try:
+            self._io.run_coro(self._proxy.stop()).result(timeout=10)
+        except Exception:  # noqa: BLE001
+            logger.debug("receiver proxy stop failed", exc_info=True)
+
+
+# -----------------------------------------------------------------------------
+# Module-level state + starters (parity: barriers.py:248-459)
+# -----------------------------------------------------------------------------
+_io_loop: Optional[IoLoop] = None
+_sender_service: Optional[SenderProxyService] = None
+_receiver_service: Optional[ReceiverProxyService] = None
+
+
+def _get_io_loop() -> IoLoop:
+    global _io_loop
+    if _io_loop is None:
+        _io_loop = IoLoop()
+    return _io_loop
+
+
+def start_receiver_proxy(
+    addresses: Dict,
+    party: str,
+    job_name: str,
+    tls_config: Optional[Dict] = None,
+    proxy_cls=None,
+    proxy_config: Optional[CrossSiloMessageConfig] = None,
+    ready_timeout_second: int = 60,
+    use_global_proxy: bool = True,
+):
+    global _receiver_service
+    if proxy_cls is None:
+        from rayfed_amd.proxy.grpc.grpc_proxy import GrpcReceiverProxy
+
+        proxy_cls = GrpcReceiverProxy
+    io = _get_io_loop()
+    proxy = proxy_cls(
+        addresses[party], party, job_name, tls_config, proxy_config
+    )
+    service = ReceiverProxyService(proxy, io)
+    service.start(ready_timeout_second=ready_timeout_second)
+    _receiver_service = service
+    _service_registry[receiver_proxy_name(job_name, use_global_proxy)] = service
+    return service
+
+
+def start_sender_proxy(
+    addresses: Dict,
+    party: str,
+    job_name: str,
+    tls_config: Optional[Dict] = None,
+    proxy_cls=None,
+    proxy_config: Optional[CrossSiloMessageConfig] = None,
+    ready_timeout_second: int = 60,
+    use_global_proxy: bool = True,
+):
+    global _sender_service
+    if proxy_cls is None:
+        from rayfed_amd.proxy.grpc.grpc_proxy import GrpcSenderProxy
+
+        proxy_cls = GrpcSenderProxy
+    io = _get_io_loop()
+    proxy = proxy_cls(addresses, party, job_name, tls_config, proxy_config)
+    service = SenderProxyService(proxy, io)
+    _sender_service = service
+    _service_registry[sender_proxy_name(job_name, use_global_proxy)] = service
+    return service
+
+
+def _cleanup_proxies():
+    """Stop proxies and the I/O loop (called from fed.shutdown)."""
+    global _sender_service, _receiver_service, _io_loop
+    if _sender_service is not None:
+        _sender_service.stop()
+        _sender_service = None
+    if _receiver_service is not None:
+        _receiver_service.stop()
+        _receiver_service = None
+    if _io_loop is not None:
+        _io_loop.stop()
+        _io_loop = None
+    _service_registry.clear()
+
+
+# -----------------------------------------------------------------------------
+# send / recv / ping (parity: barriers.py:462-523)
+# -----------------------------------------------------------------------------
+def send(
+    dest_party: str,
+    data: Any,
+    upstream_seq_id,
+    downstream_seq_id,
+    is_error: bool = False,
+) -> Future:
+    """Push ``data`` (an ObjectRef or a plain value) to ``dest_party`` under
+    the given seq ids; ack-tracked by the cleanup manager."""
+    assert _sender_service is not None, "sender proxy not started; call fed.init"
+    fut = _sender_service.send(dest_party, data, upstream_seq_id, downstream_seq_id)
+    ctx = get_global_context()
+    if ctx is not None:
+        ctx.get_cleanup_manager().push_to_sending(
+            fut, dest_party, upstream_seq_id, downstream_seq_id, is_error
+        )
+    return fut
+
+
+def recv(party: str, src_party: str, upstream_seq_id, curr_seq_id) -> ObjectRef:
+    """The receive barrier: an ObjectRef that resolves when the peer's push
+    for (upstream_seq_id, curr_seq_id) lands in the mailbox."""
+    assert _receiver_service is not None, "receiver proxy not started; call fed.init"
+    fut = _receiver_service.get_data(src_party, upstream_seq_id, curr_seq_id)
+    return ObjectRef(fut)
+
+
+def ping_others(
+    addresses: Dict[str, str], self_party: str, max_retries: int = 3600
+) -> bool:
+    """Block until every other party's receiver answers a ping
+    (parity: barriers.py:497-523)."""
+    others = [p for p in addresses if p != self_party]
+    tried = 0
+    while others and tried < max_retries:
+        logger.info("Try ping %s (attempt %d) ...", others, tried + 1)
+        still = []
+        for party in others:
+            fut = _sender_service.send(
+                party, b"data", constants.PING_SEQ_ID, constants.PING_SEQ_ID
+            )
+            try:
+                if fut.result(timeout=10) is not True:
+                    still.append(party)
+            except Exception:  # noqa: BLE001
+                still.append(party)
+        others = still
+        tried += 1
+        if others:
+            time.sleep(2)
+    if others:
+        raise RuntimeError(
+            f"Failed to ping parties {others}: receivers unreachable."
+        )
+    return True
