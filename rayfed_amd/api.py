@@ -305,7 +305,7 @@ def remote(*args, **kwargs):
             "The @fed.remote decorator must be applied to a function or a class."
         )
 
-    if len(args) == 1 and len(kwargs) == 0 and callable(args[0]):
+    if len(args) == 1 and len(kwargs) == 0:
         return _make_fed_remote(args[0])
     assert len(args) == 0 and len(kwargs) > 0, "Remote args error."
     return functools.partial(_make_fed_remote, **kwargs)

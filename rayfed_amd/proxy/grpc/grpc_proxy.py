@@ -74,6 +74,11 @@ class GrpcSenderProxy(base_proxy.SenderProxy):
                 private_key=private_key,
                 certificate_chain=cert_chain,
             )
+            # Test certs are usually issued for a DNS name while parties dial
+            # bare IPs; allow overriding the expected server name.
+            override = self._tls_config.get("target_name_override")
+            if override:
+                options = options + [("grpc.ssl_target_name_override", override)]
             channel = grpc.aio.secure_channel(address, credentials, options=options)
         else:
             channel = grpc.aio.insecure_channel(address, options=options)
@@ -177,16 +182,22 @@ class GrpcReceiverProxy(base_proxy.ReceiverProxy):
             frames.SERVICE_NAME, {"SendData": handler}
         )
         server.add_generic_rpc_handlers((generic,))
-        if self._tls_config:
-            ca_cert, private_key, cert_chain = load_cert_config(self._tls_config)
-            credentials = grpc.ssl_server_credentials(
-                [(private_key, cert_chain)],
-                root_certificates=ca_cert,
-                require_client_auth=ca_cert is not None,
-            )
-            bound = server.add_secure_port(f"[::]:{port}", credentials)
-        else:
-            bound = server.add_insecure_port(f"[::]:{port}")
+        try:
+            if self._tls_config:
+                ca_cert, private_key, cert_chain = load_cert_config(self._tls_config)
+                credentials = grpc.ssl_server_credentials(
+                    [(private_key, cert_chain)],
+                    root_certificates=ca_cert,
+                    require_client_auth=ca_cert is not None,
+                )
+                bound = server.add_secure_port(f"[::]:{port}", credentials)
+            else:
+                bound = server.add_insecure_port(f"[::]:{port}")
+        except RuntimeError as e:  # grpc ≥1.60 raises instead of returning 0
+            raise AssertionError(
+                f"Failed to listen on port {port}: it is in use ({e}). "
+                f"Choose another port in the cluster addresses."
+            ) from e
         assert str(bound) == str(port), (
             f"Failed to listen on port {port}: it is in use (got {bound}). "
             f"Choose another port in the cluster addresses."

@@ -1,0 +1,116 @@
+"""API-level units: decorator validation, init validation, kill,
+listening-address conflicts (coverage parity: reference test_api.py,
+test_listening_address.py, test_kill)."""
+import multiprocessing
+
+import pytest
+
+import rayfed_amd as fed
+from tests._util import free_ports, make_addresses, run_parties
+
+_mp = multiprocessing.get_context("fork")
+
+
+def test_remote_without_party_raises():
+    @fed.remote
+    def f():
+        return 1
+
+    with pytest.raises(ValueError, match="party"):
+        f.remote()
+
+    @fed.remote
+    class A:
+        pass
+
+    with pytest.raises(ValueError, match="party"):
+        A.remote()
+
+
+def test_remote_rejects_non_callable():
+    with pytest.raises(TypeError):
+        fed.remote(42)
+
+
+def test_init_validation():
+    with pytest.raises(AssertionError):
+        fed.init(addresses=None, party="alice")
+    with pytest.raises(AssertionError):
+        fed.init(addresses={"alice": "127.0.0.1:1"}, party=None)
+    with pytest.raises(AssertionError):
+        fed.init(addresses={"alice": "127.0.0.1:1"}, party="carol")
+    with pytest.raises(ValueError):
+        fed.init(addresses={"alice": "badaddress"}, party="alice")
+
+
+def _driver_port_in_use(party, addresses):
+    import socket
+
+    # Occupy the party's own port first: receiver bind must fail loudly.
+    s = socket.socket()
+    host, port = addresses[party].split(":")
+    s.bind((host, int(port)))
+    s.listen(1)
+    try:
+        fed.init(addresses=addresses, party=party)
+    except AssertionError:
+        import sys
+
+        sys.exit(7)  # expected path
+    finally:
+        s.close()
+
+
+def test_listening_address_in_use():
+    addresses = make_addresses(["alice"])
+    p = _mp.Process(target=_driver_port_in_use, args=("alice", addresses))
+    p.start()
+    p.join(timeout=60)
+    assert p.exitcode == 7
+
+
+def _driver_kill(party, addresses):
+    # Short transport deadline: the post-kill broadcast targets a peer that
+    # may already have exited; shutdown's drain must not wait the default 60 s.
+    fed.init(
+        addresses=addresses,
+        party=party,
+        config={"cross_silo_comm": {"timeout_in_ms": 5000}},
+    )
+
+    @fed.remote
+    class A:
+        def ping(self):
+            return "alive"
+
+    a = A.party("alice").remote()
+    r = a.ping.remote()
+    if party == "alice":
+        assert fed.get(r) == "alive"
+    fed.kill(a)
+    if party == "alice":
+        r2 = a.ping.remote()
+        with pytest.raises(Exception):
+            fed.get(r2)
+    fed.shutdown()
+
+
+def test_kill_actor():
+    run_parties(_driver_kill)
+
+
+def _driver_local_get_of_plain_ref(party, addresses):
+    fed.init(addresses=addresses, party=party)
+
+    @fed.remote
+    def f():
+        return 5
+
+    obj = f.party(party).remote()
+    ref = obj.get_ray_object_ref()
+    assert fed.get(ref) == 5  # plain ObjectRef passthrough
+    fed.shutdown()
+
+
+def test_get_plain_object_ref():
+    run_parties(_driver_local_get_of_plain_ref, parties=("alice",))

@@ -1,0 +1,51 @@
+"""Address validation + misc helpers (parity: reference without_ray_tests/test_utils.py)."""
+import pytest
+
+from rayfed_amd.utils import dict2tuple, validate_address, validate_addresses
+
+
+@pytest.mark.parametrize(
+    "address",
+    [
+        "127.0.0.1:8080",
+        "localhost:8080",
+        "my-host.example.com:443",
+        "http://example.com",
+        "https://example.com:9999",
+        "local",
+    ],
+)
+def test_valid_addresses(address):
+    validate_address(address)
+
+
+@pytest.mark.parametrize(
+    "address",
+    [
+        "127.0.0.1",          # no port
+        "127.0.0.1:0",        # port 0
+        "127.0.0.1:99999",    # port out of range
+        ":8080:extra",
+        "ftp//no",
+        12345,
+        "",
+        "-badhost-:80",
+    ],
+)
+def test_invalid_addresses(address):
+    with pytest.raises(ValueError):
+        validate_address(address)
+
+
+def test_validate_addresses_dict():
+    validate_addresses({"alice": "127.0.0.1:1", "bob": "h:2"})
+    with pytest.raises(ValueError):
+        validate_addresses({})
+    with pytest.raises(ValueError):
+        validate_addresses({"alice": "nope"})
+
+
+def test_dict2tuple():
+    assert dict2tuple(None) == []
+    assert sorted(dict2tuple({"a": 1, "b": 2})) == [("a", 1), ("b", 2)]
+    assert dict2tuple([("x", 3)]) == [("x", 3)]
