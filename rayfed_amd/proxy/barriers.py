@@ -220,9 +220,9 @@ def start_receiver_proxy(
 ):
     global _receiver_service
     if proxy_cls is None:
-        from rayfed_amd.proxy.grpc.grpc_proxy import GrpcReceiverProxy
+        from rayfed_amd.proxy.tcp.tcp_proxy import TcpReceiverProxy
 
-        proxy_cls = GrpcReceiverProxy
+        proxy_cls = TcpReceiverProxy
     io = _get_io_loop()
     proxy = proxy_cls(
         addresses[party], party, job_name, tls_config, proxy_config
@@ -246,9 +246,9 @@ def start_sender_proxy(
 ):
     global _sender_service
     if proxy_cls is None:
-        from rayfed_amd.proxy.grpc.grpc_proxy import GrpcSenderProxy
+        from rayfed_amd.proxy.tcp.tcp_proxy import TcpSenderProxy
 
-        proxy_cls = GrpcSenderProxy
+        proxy_cls = TcpSenderProxy
     io = _get_io_loop()
     proxy = proxy_cls(addresses, party, job_name, tls_config, proxy_config)
     service = SenderProxyService(proxy, io)

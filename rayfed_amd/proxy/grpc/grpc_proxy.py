@@ -27,18 +27,11 @@ from rayfed_amd._private import serialization
 from rayfed_amd.exceptions import FedRemoteError
 from rayfed_amd.ops import tensor_codec
 from rayfed_amd.proxy import base_proxy
+from rayfed_amd.proxy._mailbox import Mailbox
 from rayfed_amd.proxy.grpc import frames, grpc_options
-from rayfed_amd.utils import dict2tuple, load_cert_config
+from rayfed_amd.utils import load_cert_config
 
 logger = logging.getLogger(__name__)
-
-# Payloads larger than this are (de)serialized in the default thread-pool
-# executor so the I/O loop stays responsive for other parties' traffic.
-_OFFLOAD_BYTES = 256 * 1024
-
-
-def key_of(upstream_seq_id, downstream_seq_id) -> Tuple[str, str]:
-    return (str(upstream_seq_id), str(downstream_seq_id))
 
 
 class GrpcSenderProxy(base_proxy.SenderProxy):
@@ -119,28 +112,11 @@ class GrpcSenderProxy(base_proxy.SenderProxy):
         return True
 
     async def _encode_request(self, data, upstream_seq_id, downstream_seq_id) -> bytes:
-        header = {
-            "job": self._job_name,
-            "up": str(upstream_seq_id),
-            "down": str(downstream_seq_id),
-        }
-        if isinstance(data, FedRemoteError):
-            payload = serialization.dumps(data)
-            return frames.encode_frame(frames.KIND_ERROR, header, payload)
+        from rayfed_amd.proxy._encode import encode_request
 
-        # Tensor-aware encode: extracts torch tensors (GPU pack/CRC when a
-        # data plane is attached), falls back to a plain pickle body when the
-        # payload has none.
-        loop = asyncio.get_running_loop()
-        extras, parts = await loop.run_in_executor(
-            None, tensor_codec.encode, data, self.gpu_plane
+        return await encode_request(
+            self._job_name, data, upstream_seq_id, downstream_seq_id, self.gpu_plane
         )
-        if extras["tensors"]:
-            header.update(extras)
-            payload = b"".join(parts)
-            return frames.encode_frame(frames.KIND_TENSOR, header, payload)
-        payload = parts[0]
-        return frames.encode_frame(frames.KIND_PICKLE, header, payload)
 
     async def get_proxy_config(self, dest_party: Optional[str] = None):
         return self._proxy_config
@@ -160,12 +136,9 @@ class GrpcReceiverProxy(base_proxy.ReceiverProxy):
             proxy_config = fed_config.GrpcCrossSiloMessageConfig.from_dict(proxy_config)
         super().__init__(listening_address, party, job_name, tls_config, proxy_config)
         self._server: Optional[grpc.aio.Server] = None
-        # (up, down) -> Future resolving to (kind, header, payload)
-        self._mailbox: Dict[Tuple[str, str], asyncio.Future] = {}
-        self.gpu_plane = None
-        self.received_op_count = 0
-        self._allowed_list = (
-            proxy_config.serializing_allowed_list if proxy_config else None
+        self._mailbox = Mailbox(
+            job_name,
+            proxy_config.serializing_allowed_list if proxy_config else None,
         )
 
     # -- server ---------------------------------------------------------------
@@ -206,64 +179,31 @@ class GrpcReceiverProxy(base_proxy.ReceiverProxy):
         self._server = server
         logger.info("Receiver proxy of %s listening on %s", self._party, port)
 
+    @property
+    def gpu_plane(self):
+        return self._mailbox.gpu_plane
+
+    @gpu_plane.setter
+    def gpu_plane(self, plane):
+        self._mailbox.gpu_plane = plane
+
+    @property
+    def received_op_count(self) -> int:
+        return self._mailbox.received_op_count
+
     async def _handle_send_data(self, request: bytes, context) -> bytes:
         try:
             kind, header, payload = frames.decode_frame(request)
         except ValueError as e:
             return frames.encode_response(400, f"bad frame: {e}")
-        job_name = header.get("job", "")
-        if job_name != self._job_name:
-            logger.warning(
-                "Rejected message from job %r (serving %r)", job_name, self._job_name
-            )
-            return frames.encode_response(
-                417,
-                f"JobName mis-match: expected {self._job_name!r}, got {job_name!r}",
-            )
-        key = (header["up"], header["down"])
-        self.received_op_count += 1
-        fut = self._mailbox.get(key)
-        if fut is None or fut.done():
-            # Reader not waiting yet (or duplicate delivery): park the payload.
-            fut = asyncio.get_running_loop().create_future()
-            self._mailbox[key] = fut
-        if not fut.done():
-            fut.set_result((kind, header, bytes(payload)))
-        return frames.encode_response(200, "OK")
+        code, result = self._mailbox.deliver(kind, header, bytes(payload))
+        if code == 417:
+            logger.warning("Rejected message: %s", result)
+        return frames.encode_response(code, result)
 
     # -- consumption ----------------------------------------------------------
     async def get_data(self, src_party, upstream_seq_id, curr_seq_id):
-        key = key_of(upstream_seq_id, curr_seq_id)
-        fut = self._mailbox.get(key)
-        if fut is None:
-            fut = asyncio.get_running_loop().create_future()
-            self._mailbox[key] = fut
-        kind, header, payload = await fut
-        # Pop so the slot can be reused by a later job phase; FedObject-level
-        # caching guarantees each object is consumed once per party.
-        self._mailbox.pop(key, None)
-        loop = asyncio.get_running_loop()
-        if kind == frames.KIND_ERROR:
-            err = serialization.loads(payload, self._allowed_list)
-            raise err
-        if kind == frames.KIND_TENSOR:
-            if len(payload) >= _OFFLOAD_BYTES:
-                return await loop.run_in_executor(
-                    None,
-                    tensor_codec.decode,
-                    header,
-                    memoryview(payload),
-                    self.gpu_plane,
-                    self._allowed_list,
-                )
-            return tensor_codec.decode(
-                header, memoryview(payload), self.gpu_plane, self._allowed_list
-            )
-        if len(payload) >= _OFFLOAD_BYTES:
-            return await loop.run_in_executor(
-                None, serialization.loads, payload, self._allowed_list
-            )
-        return serialization.loads(payload, self._allowed_list)
+        return await self._mailbox.get_data(upstream_seq_id, curr_seq_id)
 
     async def stop(self):
         if self._server is not None:

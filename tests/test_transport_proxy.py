@@ -12,23 +12,34 @@ from rayfed_amd.config import GrpcCrossSiloMessageConfig
 from tests._util import make_addresses
 
 
-@pytest.fixture
-def party_env():
+@pytest.fixture(params=["tcp", "grpc"])
+def party_env(request):
+    if request.param == "tcp":
+        from rayfed_amd.proxy.tcp.tcp_proxy import TcpReceiverProxy, TcpSenderProxy
+
+        sender_cls, receiver_cls = TcpSenderProxy, TcpReceiverProxy
+    else:
+        from rayfed_amd.proxy.grpc.grpc_proxy import (
+            GrpcReceiverProxy,
+            GrpcSenderProxy,
+        )
+
+        sender_cls, receiver_cls = GrpcSenderProxy, GrpcReceiverProxy
     addrs = make_addresses(["alice"])
     init_global_context(current_party="alice", job_name="test_job")
     receiver = barriers.start_receiver_proxy(
-        addrs, "alice", job_name="test_job", proxy_config=None
+        addrs, "alice", job_name="test_job", proxy_cls=receiver_cls, proxy_config=None
     )
     sender = barriers.start_sender_proxy(
-        addrs, "alice", job_name="test_job", proxy_config=None
+        addrs, "alice", job_name="test_job", proxy_cls=sender_cls, proxy_config=None
     )
-    yield addrs, sender, receiver
+    yield request.param, addrs, sender, receiver
     clear_global_context()
     barriers._cleanup_proxies()
 
 
 def test_n_to_1_send_recv(party_env):
-    addrs, sender, receiver = party_env
+    kind, addrs, sender, receiver = party_env
     n = 10
     sends = [
         barriers.send("alice", {"i": i}, f"up{i}", f"down{i}") for i in range(n)
@@ -44,7 +55,7 @@ def test_n_to_1_send_recv(party_env):
 
 
 def test_reader_before_sender(party_env):
-    addrs, sender, receiver = party_env
+    kind, addrs, sender, receiver = party_env
     ref = barriers.recv("alice", "alice", "late", "late")
     assert not ref.done()
     barriers.send("alice", [1, 2, 3], "late", "late").result(timeout=20)
@@ -52,31 +63,53 @@ def test_reader_before_sender(party_env):
 
 
 def test_wrong_job_name_rejected_with_417(party_env):
-    """A raw frame with a mismatched job name gets code 417 and the send
-    future fails (parity: reference multi-jobs/test_ignore_other_job_msg.py)."""
-    addrs, _, _ = party_env
-    import grpc
-
+    """A raw frame with a mismatched job name gets code 417
+    (parity: reference multi-jobs/test_ignore_other_job_msg.py)."""
+    kind, addrs, _, _ = party_env
     from rayfed_amd.proxy.grpc import frames
 
-    channel = grpc.insecure_channel(addrs["alice"])
-    stub = channel.unary_unary(
-        frames.SEND_DATA_METHOD,
-        request_serializer=frames.identity_serializer,
-        response_deserializer=frames.identity_deserializer,
-    )
     raw = frames.encode_frame(
         frames.KIND_PICKLE,
         {"job": "SOME_OTHER_JOB", "up": "1", "down": "1"},
         b"payload",
     )
-    resp = frames.decode_response(stub(raw, timeout=10))
+    if kind == "grpc":
+        import grpc
+
+        channel = grpc.insecure_channel(addrs["alice"])
+        stub = channel.unary_unary(
+            frames.SEND_DATA_METHOD,
+            request_serializer=frames.identity_serializer,
+            response_deserializer=frames.identity_deserializer,
+        )
+        resp = frames.decode_response(stub(raw, timeout=10))
+        channel.close()
+    else:
+        import socket
+
+        import msgpack
+
+        host, port = addrs["alice"].rsplit(":", 1)
+        with socket.create_connection((host, int(port)), timeout=10) as s:
+            body = (1).to_bytes(8, "little") + raw
+            s.sendall(len(body).to_bytes(4, "little") + body)
+            n = int.from_bytes(_read_n(s, 4), "little")
+            resp = msgpack.unpackb(_read_n(s, n), raw=False)
     assert resp["code"] == 417
-    channel.close()
+
+
+def _read_n(sock, n):
+    buf = b""
+    while len(buf) < n:
+        chunk = sock.recv(n - len(buf))
+        if not chunk:
+            raise ConnectionError("eof")
+        buf += chunk
+    return buf
 
 
 def test_large_payload(party_env):
-    addrs, _, _ = party_env
+    kind, addrs, _, _ = party_env
     blob = b"z" * (8 * 1024 * 1024)
     barriers.send("alice", blob, "big", "big").result(timeout=30)
     ref = barriers.recv("alice", "alice", "big", "big")
@@ -85,7 +118,7 @@ def test_large_payload(party_env):
 
 def test_tensor_payload_over_wire(party_env):
     torch = pytest.importorskip("torch")
-    addrs, _, _ = party_env
+    kind, addrs, _, _ = party_env
     t = torch.randn(1000, dtype=torch.float32)
     barriers.send("alice", {"w": t}, "tens", "tens").result(timeout=30)
     out = barriers.recv("alice", "alice", "tens", "tens").result(timeout=30)
