@@ -1,0 +1,616 @@
+// rayfed_amd HIP extension — MI355X (gfx950) data-plane kernels.
+//
+// Replaces the reference's CPU cloudpickle payload path
+// (/root/reference/fed/proxy/grpc/grpc_proxy.py:202) with device-side
+// pack/CRC and aggregation:
+//
+//   * crc32       — zlib-compatible CRC32 of a device byte buffer, fully
+//                   parallel: per-thread slice CRCs (slice-by-8, tables in
+//                   LDS) + GF(2) shift-matrix combine in-kernel, finalized
+//                   by a 1-thread kernel.  Exact for any length.
+//   * pack_crc    — fused flatten-copy + CRC32 (one HBM read pass) into the
+//                   device staging buffer that hipMemcpyAsync then DMAs to
+//                   pinned host memory.
+//   * pack_fp8 /  — wire compression: bf16/f32 -> OCP fp8 e4m3 cast fused
+//     unpack_fp8    with CRC (half the bytes on the wire), and the inverse.
+//   * fedavg_reduce_ / masked_add_ — weighted gradient aggregation for
+//     intra-party FedAvg (bf16/f16/f32, fp32 accumulation, vectorized
+//     16-byte loads per lane — CDNA4 guide Appendix B: element-wise ops are
+//     HBM-bound, always vectorize).
+//
+// Wavefront size is 64 on CDNA4 (not 32); reductions below use width-64
+// shuffles.  Compiled for gfx950 only — no CUDA shims, no multi-arch.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+#include "crc32_math.h"
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));    \
+  } while (0)
+
+namespace {
+
+constexpr int kWave = 64;              // CDNA4 wavefront
+constexpr int kBlock = 256;            // threads per workgroup
+constexpr uint32_t kSubLen = 4096;     // bytes CRC'd per thread (uniform)
+constexpr int kMaxPow = 24;            // M_sub^(2^j), j < kMaxPow
+
+// ---------------------------------------------------------------------------
+// device-side GF(2) apply
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint32_t gf2_apply(const uint32_t* m, uint32_t v) {
+  uint32_t s = 0;
+#pragma unroll
+  for (int i = 0; i < 32; ++i) {
+    s ^= (m[i] & (0u - ((v >> i) & 1u)));
+  }
+  return s;
+}
+
+// slice-by-8 step over 8 bytes (init-0 linear CRC), tables in LDS.
+__device__ __forceinline__ uint32_t crc_step8(const uint32_t* t, uint32_t crc,
+                                              uint32_t lo, uint32_t hi) {
+  lo ^= crc;
+  return t[7 * 256 + (lo & 0xFFu)] ^ t[6 * 256 + ((lo >> 8) & 0xFFu)] ^
+         t[5 * 256 + ((lo >> 16) & 0xFFu)] ^ t[4 * 256 + (lo >> 24)] ^
+         t[3 * 256 + (hi & 0xFFu)] ^ t[2 * 256 + ((hi >> 8) & 0xFFu)] ^
+         t[1 * 256 + ((hi >> 16) & 0xFFu)] ^ t[0 * 256 + (hi >> 24)];
+}
+
+struct CrcShared {
+  uint32_t tables[8 * 256];
+  uint32_t pows[kMaxPow * 32];
+};
+
+__device__ __forceinline__ void load_crc_shared(CrcShared& sh,
+                                                const uint32_t* g_tables,
+                                                const uint32_t* g_pows) {
+  for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x)
+    sh.tables[i] = g_tables[i];
+  for (int i = threadIdx.x; i < kMaxPow * 32; i += blockDim.x)
+    sh.pows[i] = g_pows[i];
+  __syncthreads();
+}
+
+// xor-reduce `v` across the block, atomically xor into *out from one lane.
+__device__ __forceinline__ void block_xor_out(uint32_t v, uint32_t* out,
+                                              uint32_t* lds_scratch) {
+#pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1) v ^= __shfl_down(v, off, kWave);
+  const int wid = threadIdx.x / kWave;
+  if ((threadIdx.x & (kWave - 1)) == 0) lds_scratch[wid] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t acc = 0;
+    for (int w = 0; w < (int)(blockDim.x / kWave); ++w) acc ^= lds_scratch[w];
+    if (acc) atomicXor(out, acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// CRC32 kernel (optionally fused with a flatten copy src->dst)
+//
+// Thread t < T-1 CRCs bytes [t*kSubLen, (t+1)*kSubLen); the last thread takes
+// the tail.  Full threads apply M_sub^(T-2-t) on-device (binary powers in
+// LDS); the shared tail shift + init/final constant are applied by
+// crc_finalize_kernel.  out[0] = xor of full-thread terms, out[1] = L(tail).
+// ---------------------------------------------------------------------------
+template <bool kPack>
+__global__ void crc32_kernel(const uint8_t* __restrict__ data,
+                             uint8_t* __restrict__ dst,
+                             unsigned long long n, uint32_t T,
+                             const uint32_t* __restrict__ g_tables,
+                             const uint32_t* __restrict__ g_pows,
+                             uint32_t* __restrict__ out) {
+  __shared__ CrcShared sh;
+  __shared__ uint32_t lds_scratch[kBlock / kWave];
+  load_crc_shared(sh, g_tables, g_pows);
+
+  const uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t contrib_full = 0, contrib_tail = 0;
+  if (t < T) {
+    const unsigned long long off = (unsigned long long)t * kSubLen;
+    const uint32_t len = (t == T - 1) ? (uint32_t)(n - off) : kSubLen;
+    uint32_t crc = 0;
+    const uint8_t* p = data + off;
+    uint8_t* q = kPack ? dst + off : nullptr;
+    uint32_t i = 0;
+    if ((((uintptr_t)p) & 15u) == 0) {
+      for (; i + 16 <= len; i += 16) {
+        const uint4 v = *reinterpret_cast<const uint4*>(p + i);
+        if (kPack) *reinterpret_cast<uint4*>(q + i) = v;
+        crc = crc_step8(sh.tables, crc, v.x, v.y);
+        crc = crc_step8(sh.tables, crc, v.z, v.w);
+      }
+    }
+    for (; i < len; ++i) {
+      const uint8_t b = p[i];
+      if (kPack) q[i] = b;
+      crc = (crc >> 8) ^ sh.tables[(crc ^ b) & 0xFFu];
+    }
+    if (t == T - 1) {
+      contrib_tail = crc;
+    } else {
+      uint32_t k = T - 2 - t;
+      int j = 0;
+      while (k) {
+        if (k & 1u) crc = gf2_apply(&sh.pows[j * 32], crc);
+        k >>= 1;
+        ++j;
+      }
+      contrib_full = crc;
+    }
+  }
+  block_xor_out(contrib_full, &out[0], lds_scratch);
+  __syncthreads();
+  block_xor_out(contrib_tail, &out[1], lds_scratch);
+}
+
+struct TailParam {
+  uint32_t m[32];       // shift matrix for the tail length
+  uint32_t final_xor;   // shift_len(0xFFFFFFFF) ^ 0xFFFFFFFF
+};
+
+__global__ void crc_finalize_kernel(uint32_t* out, TailParam p) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    uint32_t s = 0;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) s ^= (p.m[i] & (0u - ((out[0] >> i) & 1u)));
+    out[2] = s ^ out[1] ^ p.final_xor;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fp8 wire cast (OCP e4m3), fused with CRC over the produced bytes
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float bf16_to_f32(uint16_t h) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.u = (uint32_t)h << 16;
+  return c.f;
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.f = f;
+  // round-to-nearest-even
+  uint32_t r = c.u + 0x7FFFu + ((c.u >> 16) & 1u);
+  return (uint16_t)(r >> 16);
+}
+
+__device__ __forceinline__ uint8_t f32_to_fp8(float f) {
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
+}
+
+__device__ __forceinline__ float fp8_to_f32(uint8_t b) {
+  __hip_fp8_e4m3 v;
+  v.__x = b;
+  return (float)v;
+}
+
+// src: bf16 elements; dst: fp8 bytes.  Thread t produces kSubLen output
+// bytes (= kSubLen input bf16 elements); same combine scheme as crc32_kernel.
+__global__ void pack_fp8_kernel(const uint16_t* __restrict__ src,
+                                uint8_t* __restrict__ dst,
+                                unsigned long long n_elems, uint32_t T,
+                                const uint32_t* __restrict__ g_tables,
+                                const uint32_t* __restrict__ g_pows,
+                                uint32_t* __restrict__ out) {
+  __shared__ CrcShared sh;
+  __shared__ uint32_t lds_scratch[kBlock / kWave];
+  load_crc_shared(sh, g_tables, g_pows);
+
+  const uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t contrib_full = 0, contrib_tail = 0;
+  if (t < T) {
+    const unsigned long long off = (unsigned long long)t * kSubLen;
+    const uint32_t len =
+        (t == T - 1) ? (uint32_t)(n_elems - off) : kSubLen;  // elements
+    uint32_t crc = 0;
+    const uint16_t* p = src + off;
+    uint8_t* q = dst + off;
+    uint32_t i = 0;
+    for (; i + 8 <= len; i += 8) {
+      const uint4 v = *reinterpret_cast<const uint4*>(p + i);  // 8 bf16
+      uint32_t lo = 0, hi = 0;
+      lo |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.x & 0xFFFFu)));
+      lo |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.x >> 16))) << 8;
+      lo |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.y & 0xFFFFu))) << 16;
+      lo |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.y >> 16))) << 24;
+      hi |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.z & 0xFFFFu)));
+      hi |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.z >> 16))) << 8;
+      hi |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.w & 0xFFFFu))) << 16;
+      hi |= (uint32_t)f32_to_fp8(bf16_to_f32((uint16_t)(v.w >> 16))) << 24;
+      *reinterpret_cast<uint2*>(q + i) = make_uint2(lo, hi);
+      crc = crc_step8(sh.tables, crc, lo, hi);
+    }
+    for (; i < len; ++i) {
+      const uint8_t b = f32_to_fp8(bf16_to_f32(p[i]));
+      q[i] = b;
+      crc = (crc >> 8) ^ sh.tables[(crc ^ b) & 0xFFu];
+    }
+    if (t == T - 1) {
+      contrib_tail = crc;
+    } else {
+      uint32_t k = T - 2 - t;
+      int j = 0;
+      while (k) {
+        if (k & 1u) crc = gf2_apply(&sh.pows[j * 32], crc);
+        k >>= 1;
+        ++j;
+      }
+      contrib_full = crc;
+    }
+  }
+  block_xor_out(contrib_full, &out[0], lds_scratch);
+  __syncthreads();
+  block_xor_out(contrib_tail, &out[1], lds_scratch);
+}
+
+__global__ void unpack_fp8_kernel(const uint8_t* __restrict__ src,
+                                  uint16_t* __restrict__ dst,
+                                  unsigned long long n_elems) {
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x * 8ull;
+  for (unsigned long long base =
+           ((unsigned long long)blockIdx.x * blockDim.x + threadIdx.x) * 8ull;
+       base < n_elems; base += stride) {
+    if (base + 8 <= n_elems) {
+      const uint2 v = *reinterpret_cast<const uint2*>(src + base);
+      uint4 o;
+      o.x = (uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.x))) |
+            ((uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.x >> 8))) << 16);
+      o.y = (uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.x >> 16))) |
+            ((uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.x >> 24))) << 16);
+      o.z = (uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.y))) |
+            ((uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.y >> 8))) << 16);
+      o.w = (uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.y >> 16))) |
+            ((uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(v.y >> 24))) << 16);
+      *reinterpret_cast<uint4*>(dst + base) = o;
+    } else {
+      for (unsigned long long i = base; i < n_elems; ++i)
+        dst[i] = f32_to_bf16(fp8_to_f32(src[i]));
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FedAvg weighted reduce: out = sum_k w_k * in_k   (fp32 accumulation)
+// ---------------------------------------------------------------------------
+constexpr int kMaxInputs = 16;
+
+struct ReduceArgs {
+  const void* in[kMaxInputs];
+  float w[kMaxInputs];
+  int k;
+};
+
+template <typename T>
+__device__ __forceinline__ float to_f32(T v);
+template <>
+__device__ __forceinline__ float to_f32<uint16_t>(uint16_t v) {
+  return bf16_to_f32(v);
+}
+template <>
+__device__ __forceinline__ float to_f32<__half>(__half v) {
+  return __half2float(v);
+}
+template <>
+__device__ __forceinline__ float to_f32<float>(float v) {
+  return v;
+}
+
+template <typename T>
+__device__ __forceinline__ T from_f32(float v);
+template <>
+__device__ __forceinline__ uint16_t from_f32<uint16_t>(float v) {
+  return f32_to_bf16(v);
+}
+template <>
+__device__ __forceinline__ __half from_f32<__half>(float v) {
+  return __float2half(v);
+}
+template <>
+__device__ __forceinline__ float from_f32<float>(float v) {
+  return v;
+}
+
+// VEC elements per lane per iteration, sized to 16-byte loads.
+template <typename T, int VEC>
+__global__ void fedavg_reduce_kernel(T* __restrict__ out, ReduceArgs args,
+                                     unsigned long long n) {
+  const unsigned long long stride =
+      (unsigned long long)gridDim.x * blockDim.x * VEC;
+  for (unsigned long long base =
+           ((unsigned long long)blockIdx.x * blockDim.x + threadIdx.x) * VEC;
+       base < n; base += stride) {
+    float acc[VEC];
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) acc[v] = 0.0f;
+    const bool full = base + VEC <= n;
+    for (int j = 0; j < args.k; ++j) {
+      const T* in = reinterpret_cast<const T*>(args.in[j]);
+      const float w = args.w[j];
+      if (full) {
+        // One 16-byte load per input per iteration.
+        using VecT = uint4;
+        const VecT raw = *reinterpret_cast<const VecT*>(in + base);
+        const T* e = reinterpret_cast<const T*>(&raw);
+#pragma unroll
+        for (int v = 0; v < VEC; ++v) acc[v] += w * to_f32<T>(e[v]);
+      } else {
+        for (unsigned long long i = base; i < n; ++i)
+          acc[i - base] += w * to_f32<T>(in[i]);
+      }
+    }
+    if (full) {
+      T tmp[VEC];
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) tmp[v] = from_f32<T>(acc[v]);
+      *reinterpret_cast<uint4*>(out + base) = *reinterpret_cast<uint4*>(tmp);
+    } else {
+      for (unsigned long long i = base; i < n; ++i)
+        out[i] = from_f32<T>(acc[i - base]);
+    }
+  }
+}
+
+// dst += mask ? src : 0   (secure-aggregation style mask-add)
+template <typename T>
+__global__ void masked_add_kernel(T* __restrict__ dst,
+                                  const T* __restrict__ src,
+                                  const uint8_t* __restrict__ mask,
+                                  unsigned long long n) {
+  const unsigned long long stride = (unsigned long long)gridDim.x * blockDim.x;
+  for (unsigned long long i =
+           (unsigned long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    if (mask[i])
+      dst[i] = from_f32<T>(to_f32<T>(dst[i]) + to_f32<T>(src[i]));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host-side state: tables + power matrices on device (uploaded once)
+// ---------------------------------------------------------------------------
+struct DeviceConsts {
+  uint32_t* tables = nullptr;  // 8*256
+  uint32_t* pows = nullptr;    // kMaxPow*32
+};
+
+DeviceConsts& get_device_consts() {
+  static DeviceConsts consts;
+  static std::once_flag flag;
+  std::call_once(flag, [] {
+    std::vector<uint32_t> tables(8 * 256);
+    rayfed_crc::make_slice8_tables(tables.data());
+    std::vector<uint32_t> pows(kMaxPow * 32);
+    // pows[j] = shift matrix for kSubLen * 2^j zero bytes
+    uint32_t m[32];
+    rayfed_crc::shift_matrix(m, kSubLen);
+    std::memcpy(&pows[0], m, sizeof(m));
+    for (int j = 1; j < kMaxPow; ++j) {
+      rayfed_crc::gf2_square(&pows[j * 32], &pows[(j - 1) * 32]);
+    }
+    HIP_CHECK(hipMalloc(&consts.tables, tables.size() * 4));
+    HIP_CHECK(hipMalloc(&consts.pows, pows.size() * 4));
+    HIP_CHECK(hipMemcpy(consts.tables, tables.data(), tables.size() * 4,
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(consts.pows, pows.data(), pows.size() * 4,
+                        hipMemcpyHostToDevice));
+  });
+  return consts;
+}
+
+TailParam make_tail_param(unsigned long long n, uint32_t T) {
+  TailParam p;
+  const unsigned long long tail =
+      (T == 0) ? 0 : n - (unsigned long long)(T - 1) * kSubLen;
+  rayfed_crc::shift_matrix(p.m, tail);
+  p.final_xor = rayfed_crc::shift_apply(n, 0xFFFFFFFFu) ^ 0xFFFFFFFFu;
+  return p;
+}
+
+inline uint32_t n_threads_for(unsigned long long nbytes) {
+  return (uint32_t)((nbytes + kSubLen - 1) / kSubLen);
+}
+
+// ---------------------------------------------------------------------------
+// torch bindings
+// ---------------------------------------------------------------------------
+torch::Tensor crc32_async(torch::Tensor bytes) {
+  TORCH_CHECK(bytes.is_cuda() && bytes.dtype() == torch::kUInt8 &&
+                  bytes.is_contiguous(),
+              "crc32 expects a contiguous CUDA uint8 tensor");
+  const unsigned long long n = bytes.numel();
+  auto out = torch::zeros(
+      {3}, torch::dtype(torch::kInt32).device(bytes.device()));
+  if (n == 0) return out;  // crc32("") == 0 — out[2] already 0
+  auto& consts = get_device_consts();
+  const uint32_t T = n_threads_for(n);
+  const uint32_t blocks = (T + kBlock - 1) / kBlock;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(crc32_kernel<false>, dim3(blocks), dim3(kBlock), 0,
+                     stream, bytes.data_ptr<uint8_t>(), nullptr, n, T,
+                     consts.tables, consts.pows,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
+  hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
+                     make_tail_param(n, T));
+  return out;
+}
+
+int64_t crc32_sync(torch::Tensor bytes) {
+  auto out = crc32_async(bytes);
+  return (int64_t)(uint32_t)out[2].item<int32_t>();
+}
+
+torch::Tensor pack_crc_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kUInt8 &&
+                  src.is_contiguous(),
+              "pack_crc expects contiguous CUDA uint8 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kUInt8 &&
+                  dst.is_contiguous() && dst.numel() >= src.numel(),
+              "pack_crc dst too small");
+  const unsigned long long n = src.numel();
+  auto out = torch::zeros(
+      {3}, torch::dtype(torch::kInt32).device(src.device()));
+  if (n == 0) return out;
+  auto& consts = get_device_consts();
+  const uint32_t T = n_threads_for(n);
+  const uint32_t blocks = (T + kBlock - 1) / kBlock;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(crc32_kernel<true>, dim3(blocks), dim3(kBlock), 0,
+                     stream, src.data_ptr<uint8_t>(), dst.data_ptr<uint8_t>(),
+                     n, T, consts.tables, consts.pows,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
+  hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
+                     make_tail_param(n, T));
+  return out;
+}
+
+torch::Tensor pack_fp8_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kBFloat16 &&
+                  src.is_contiguous(),
+              "pack_fp8 expects contiguous CUDA bf16 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kUInt8 &&
+                  dst.numel() >= src.numel(),
+              "pack_fp8 dst too small");
+  const unsigned long long n = src.numel();
+  auto out = torch::zeros(
+      {3}, torch::dtype(torch::kInt32).device(src.device()));
+  if (n == 0) return out;
+  auto& consts = get_device_consts();
+  const uint32_t T = n_threads_for(n);
+  const uint32_t blocks = (T + kBlock - 1) / kBlock;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(pack_fp8_kernel, dim3(blocks), dim3(kBlock), 0, stream,
+                     reinterpret_cast<const uint16_t*>(src.data_ptr()),
+                     dst.data_ptr<uint8_t>(), n, T, consts.tables, consts.pows,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
+  hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
+                     reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
+                     make_tail_param(n, T));
+  return out;
+}
+
+void unpack_fp8_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kUInt8 &&
+                  src.is_contiguous(),
+              "unpack_fp8 expects contiguous CUDA uint8 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kBFloat16 &&
+                  dst.is_contiguous() && dst.numel() == src.numel(),
+              "unpack_fp8 dst mismatch");
+  const unsigned long long n = src.numel();
+  if (n == 0) return;
+  const uint32_t blocks =
+      std::min<unsigned long long>(4096ull, (n / 8 + kBlock - 1) / kBlock + 1);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(unpack_fp8_kernel, dim3(blocks), dim3(kBlock), 0, stream,
+                     src.data_ptr<uint8_t>(),
+                     reinterpret_cast<uint16_t*>(dst.data_ptr()), n);
+}
+
+void fedavg_reduce_(torch::Tensor out, std::vector<torch::Tensor> inputs,
+                    std::vector<double> weights) {
+  TORCH_CHECK(!inputs.empty() && inputs.size() <= kMaxInputs,
+              "fedavg_reduce_: 1..16 inputs");
+  TORCH_CHECK(weights.size() == inputs.size(), "weights/inputs mismatch");
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "out must be CUDA contiguous");
+  const unsigned long long n = out.numel();
+  ReduceArgs args;
+  args.k = (int)inputs.size();
+  for (int j = 0; j < args.k; ++j) {
+    TORCH_CHECK(inputs[j].is_cuda() && inputs[j].is_contiguous() &&
+                    inputs[j].numel() == (long long)n &&
+                    inputs[j].dtype() == out.dtype(),
+                "input ", j, " mismatch");
+    args.in[j] = inputs[j].data_ptr();
+    args.w[j] = (float)weights[j];
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  // >= 2048 workgroups to fill 256 CUs across 8 XCDs (guide §1).
+  const int blocks = 4096;
+  if (out.dtype() == torch::kBFloat16) {
+    hipLaunchKernelGGL((fedavg_reduce_kernel<uint16_t, 8>), dim3(blocks),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<uint16_t*>(out.data_ptr()), args, n);
+  } else if (out.dtype() == torch::kHalf) {
+    hipLaunchKernelGGL((fedavg_reduce_kernel<__half, 8>), dim3(blocks),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<__half*>(out.data_ptr()), args, n);
+  } else if (out.dtype() == torch::kFloat) {
+    hipLaunchKernelGGL((fedavg_reduce_kernel<float, 4>), dim3(blocks),
+                       dim3(kBlock), 0, stream, out.data_ptr<float>(), args, n);
+  } else {
+    TORCH_CHECK(false, "fedavg_reduce_: unsupported dtype ", out.dtype());
+  }
+}
+
+void masked_add_(torch::Tensor dst, torch::Tensor src, torch::Tensor mask) {
+  TORCH_CHECK(dst.is_cuda() && dst.is_contiguous() && src.is_contiguous() &&
+                  mask.is_contiguous(),
+              "masked_add_: contiguous CUDA tensors required");
+  TORCH_CHECK(src.numel() == dst.numel() && mask.numel() == dst.numel(),
+              "masked_add_: size mismatch");
+  TORCH_CHECK(mask.dtype() == torch::kUInt8 || mask.dtype() == torch::kBool,
+              "mask must be uint8/bool");
+  const unsigned long long n = dst.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  const int blocks = 4096;
+  if (dst.dtype() == torch::kBFloat16) {
+    hipLaunchKernelGGL((masked_add_kernel<uint16_t>), dim3(blocks),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<uint16_t*>(dst.data_ptr()),
+                       reinterpret_cast<const uint16_t*>(src.data_ptr()),
+                       static_cast<const uint8_t*>(mask.data_ptr()), n);
+  } else if (dst.dtype() == torch::kFloat) {
+    hipLaunchKernelGGL((masked_add_kernel<float>), dim3(blocks), dim3(kBlock),
+                       0, stream, dst.data_ptr<float>(), src.data_ptr<float>(),
+                       static_cast<const uint8_t*>(mask.data_ptr()), n);
+  } else {
+    TORCH_CHECK(false, "masked_add_: unsupported dtype ", dst.dtype());
+  }
+}
+
+int64_t crc32_combine_py(int64_t crc1, int64_t crc2, int64_t len2) {
+  return (int64_t)rayfed_crc::crc32_combine((uint32_t)crc1, (uint32_t)crc2,
+                                            (uint64_t)len2);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "rayfed_amd MI355X data-plane kernels (gfx950)";
+  m.def("crc32_async", &crc32_async,
+        "CRC32 of a device uint8 tensor -> int32[3] device tensor "
+        "(out[2] = finalized CRC)");
+  m.def("crc32", &crc32_sync, "CRC32 of a device uint8 tensor (synchronizes)");
+  m.def("pack_crc_async", &pack_crc_async,
+        "fused copy src->dst + CRC32; returns int32[3] device tensor");
+  m.def("pack_fp8_async", &pack_fp8_async,
+        "fused bf16 -> OCP fp8 e4m3 cast + CRC32 of the fp8 bytes");
+  m.def("unpack_fp8_async", &unpack_fp8_async, "fp8 e4m3 bytes -> bf16");
+  m.def("fedavg_reduce_", &fedavg_reduce_,
+        "out = sum_k w_k * in_k (bf16/f16/f32, fp32 accumulation)");
+  m.def("masked_add_", &masked_add_, "dst += mask ? src : 0");
+  m.def("crc32_combine", &crc32_combine_py, "zlib-style CRC combine");
+}

@@ -140,6 +140,9 @@ class GpuDataPlaneConfig:
     verify_crc: bool = True
     # Receive tensors straight back onto the GPU of the consuming party.
     place_on_gpu: bool = True
+    # Optional lossy wire compression for bf16 tensors: 'fp8e4m3' casts to
+    # OCP fp8 on the wire (fused HIP cast+CRC kernel), halving bytes.
+    wire_dtype: Optional[str] = None
 
     @classmethod
     def from_dict(cls, data: Optional[Dict]) -> "GpuDataPlaneConfig":

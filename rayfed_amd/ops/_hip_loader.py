@@ -22,6 +22,8 @@ def load():
     with _lock:
         if _ext is None:
             try:
+                import torch  # noqa: F401 — loads libc10/libtorch first
+
                 _ext = importlib.import_module("rayfed_amd._hip")
             except ImportError as e:
                 raise ImportError(

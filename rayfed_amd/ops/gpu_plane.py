@@ -4,17 +4,21 @@ MI355X-native replacement for the reference's CPU pickle of payloads
 (/root/reference/fed/proxy/grpc/grpc_proxy.py:202).  Pipeline per tensor
 (SURVEY.md §2.3 / §7 step 3):
 
-  send:  HIP pack+CRC32 kernel on a side stream → chunked hipMemcpyAsync
-         D2H into pinned staging → bytes handed to the gRPC frame
+  send:  HIP CRC32 kernel on a dedicated stream (one HBM read pass,
+         parallel slice CRC + GF(2) combine — csrc/rayfed_hip.hip)
+         ∥ hipMemcpyAsync D2H into pinned staging on the copy stream
+         → bytes handed to the transport frame
   recv:  frame bytes → pinned staging → hipMemcpyAsync H2D on a side
-         stream → HIP CRC32 verify kernel → device tensor
+         stream → HIP CRC32 verify on the device tensor
 
-The CRC32 and pack kernels live in the in-tree HIP extension
-(``csrc/pack_crc.hip`` → ``rayfed_amd._hip``); when the extension or a GPU is
-absent this module is not instantiated and the CPU fallback in
-``tensor_codec`` applies.  On a GPU box the HIP extension is REQUIRED — we
-fail loudly rather than fall back silently (so a "GPU test" can never pass on
-an eager CPU path by accident).
+Optional wire compression (``wire_dtype='fp8e4m3'``): bf16 tensors are cast
+to OCP fp8 e4m3 by the fused pack_fp8 kernel (cast + CRC in one pass) and
+expanded back to bf16 on the receiver — half the bytes on the wire, opt-in
+because it is lossy.
+
+The HIP extension is REQUIRED on a GPU box — this module raises instead of
+falling back to an eager path, so a GPU test can never silently pass on CPU
+code.
 """
 from __future__ import annotations
 
@@ -33,7 +37,6 @@ except ImportError:  # pragma: no cover
 
 
 def _load_hip_ext():
-    """Import the in-tree HIP extension; raise ImportError when missing."""
     from rayfed_amd.ops import _hip_loader
 
     return _hip_loader.load()
@@ -46,91 +49,137 @@ class GpuDataPlane:
         if torch is None or not torch.cuda.is_available():
             raise RuntimeError("GpuDataPlane requires a visible HIP device")
         self.config = config
-        self.device = torch.device("cuda", device or torch.cuda.current_device())
-        # HIP extension is mandatory on-GPU: loud failure over silent fallback.
-        self._ext = _load_hip_ext()
-        # Side streams so pack/copy overlap compute and each other.
-        self._d2h_stream = torch.cuda.Stream(device=self.device)
-        self._h2d_stream = torch.cuda.Stream(device=self.device)
+        self.device = torch.device(
+            "cuda", device if device is not None else torch.cuda.current_device()
+        )
+        self._ext = _load_hip_ext()  # loud failure over silent fallback
+        # Dedicated streams: CRC kernel and DMA copy overlap each other and
+        # the main compute stream (hipEvent-gated, no global syncs).
+        self._crc_stream = torch.cuda.Stream(device=self.device)
+        self._copy_stream = torch.cuda.Stream(device=self.device)
         self._lock = threading.Lock()
         self._pinned: List[torch.Tensor] = []
+        self._dev_staging: List[torch.Tensor] = []
 
-    # -- pinned staging pool --------------------------------------------------
-    def _get_pinned(self, nbytes: int) -> torch.Tensor:
+    # -- buffer pools ---------------------------------------------------------
+    def _get_buf(self, pool: List, nbytes: int, pin: bool) -> "torch.Tensor":
         with self._lock:
-            for i, buf in enumerate(self._pinned):
+            for i, buf in enumerate(pool):
                 if buf.numel() >= nbytes:
-                    return self._pinned.pop(i)
-        return torch.empty(
-            max(nbytes, self.config.chunk_bytes), dtype=torch.uint8, pin_memory=True
-        )
+                    return pool.pop(i)
+        if pin:
+            return torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+        return torch.empty(nbytes, dtype=torch.uint8, device=self.device)
 
-    def _put_pinned(self, buf: torch.Tensor) -> None:
+    def _put_buf(self, pool: List, buf: "torch.Tensor") -> None:
         with self._lock:
-            if len(self._pinned) < self.config.staging_buffers:
-                self._pinned.append(buf)
+            if len(pool) < self.config.staging_buffers:
+                pool.append(buf)
+                return
+        del buf
 
     # -- send path ------------------------------------------------------------
-    def pack_to_host(self, t: "torch.Tensor") -> Tuple[memoryview, Optional[int]]:
-        """Flatten ``t`` to raw bytes in pinned host memory; return
-        (bytes view, crc32|None).  CRC is computed on-device by the HIP
-        kernel, overlapped with the D2H copy on the side stream."""
+    def pack_to_host(self, t: "torch.Tensor") -> Tuple[bytes, Optional[int]]:
+        """Flatten ``t`` to raw wire bytes in pinned host memory.
+
+        Returns (bytes-like view over pinned staging, crc32|None).  The CRC
+        kernel and the D2H DMA run on separate streams off the producing
+        stream's event, overlapping each other.
+        """
         t = t.detach()
         if not t.is_contiguous():
             t = t.contiguous()
         nbytes = t.numel() * t.element_size()
-        flat = t.view(-1).view(torch.uint8) if nbytes else t.new_empty(0, dtype=torch.uint8)
-        staging = self._get_pinned(nbytes)
+        if nbytes == 0:
+            return b"", (0 if self.config.verify_crc else None)
+        flat = t.view(-1).view(torch.uint8)
+
+        wire_fp8 = (
+            self.config.wire_dtype == "fp8e4m3" and t.dtype == torch.bfloat16
+        )
+        produced = torch.cuda.current_stream(self.device).record_event()
+        crc_out = None
+        if wire_fp8:
+            wire_bytes = t.numel()
+            dev_staging = self._get_buf(self._dev_staging, wire_bytes, pin=False)
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+                crc_out = self._ext.pack_fp8_async(
+                    t.view(-1), dev_staging[:wire_bytes]
+                )
+                packed = self._crc_stream.record_event()
+            pinned = self._get_buf(self._pinned, wire_bytes, pin=True)
+            with torch.cuda.stream(self._copy_stream):
+                self._copy_stream.wait_event(packed)
+                pinned[:wire_bytes].copy_(dev_staging[:wire_bytes], non_blocking=True)
+                done = self._copy_stream.record_event()
+            done.synchronize()
+            self._put_buf(self._dev_staging, dev_staging)
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF if self.config.verify_crc else None
+            out = pinned[:wire_bytes].numpy().tobytes()
+            self._put_buf(self._pinned, pinned)
+            return out, crc
+
+        # Same-dtype wire: CRC pass and D2H DMA read `flat` concurrently.
+        pinned = self._get_buf(self._pinned, nbytes, pin=True)
+        if self.config.verify_crc:
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+                crc_out = self._ext.crc32_async(flat)
+                crc_done = self._crc_stream.record_event()
+        with torch.cuda.stream(self._copy_stream):
+            self._copy_stream.wait_event(produced)
+            pinned[:nbytes].copy_(flat, non_blocking=True)
+            copy_done = self._copy_stream.record_event()
+        copy_done.synchronize()
         crc = None
-        with torch.cuda.stream(self._d2h_stream):
-            self._d2h_stream.wait_stream(torch.cuda.current_stream(self.device))
-            if nbytes:
-                staging[:nbytes].copy_(flat, non_blocking=True)
-            if self.config.verify_crc and nbytes:
-                crc_t = self._ext.crc32(flat)
-            done = torch.cuda.Event()
-            done.record(self._d2h_stream)
-        done.synchronize()
-        if self.config.verify_crc and nbytes:
-            crc = int(crc_t.item()) & 0xFFFFFFFF
-        out = staging[:nbytes].numpy().data
-        # NOTE: the staging buffer is handed to the frame encoder as a view;
-        # it returns to the pool only after the bytes are copied onto the
-        # wire (frames.encode_frame materializes with bytes()).
-        self._put_pinned(staging)
+        if self.config.verify_crc:
+            crc_done.synchronize()
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
+        out = pinned[:nbytes].numpy().tobytes()
+        self._put_buf(self._pinned, pinned)
         return out, crc
 
     # -- recv path ------------------------------------------------------------
     def unpack_from_host(
         self,
-        raw: memoryview,
+        raw,
         dtype: "torch.dtype",
         shape: List[int],
         crc_expect: Optional[int],
+        wire_dtype: Optional[str] = None,
     ) -> "torch.Tensor":
         nbytes = len(raw)
-        staging = self._get_pinned(nbytes)
-        if nbytes:
-            staging[:nbytes].copy_(
-                torch.frombuffer(bytearray(raw), dtype=torch.uint8)
-            )
         out = torch.empty(shape, dtype=dtype, device=self.device)
-        with torch.cuda.stream(self._h2d_stream):
-            flat = out.view(-1).view(torch.uint8) if nbytes else None
-            if nbytes:
-                flat.copy_(staging[:nbytes], non_blocking=True)
+        if nbytes == 0:
+            return out
+        pinned = self._get_buf(self._pinned, nbytes, pin=True)
+        pinned[:nbytes].numpy()[:] = memoryview(raw).cast("B")
+
+        wire_fp8 = wire_dtype == "fp8e4m3"
+        with torch.cuda.stream(self._copy_stream):
+            if wire_fp8:
+                dev_staging = self._get_buf(self._dev_staging, nbytes, pin=False)
+                dev_staging[:nbytes].copy_(pinned[:nbytes], non_blocking=True)
                 if self.config.verify_crc and crc_expect is not None:
-                    crc_t = self._ext.crc32(flat)
-            done = torch.cuda.Event()
-            done.record(self._h2d_stream)
+                    crc_out = self._ext.crc32_async(dev_staging[:nbytes])
+                self._ext.unpack_fp8_async(dev_staging[:nbytes], out.view(-1))
+            else:
+                flat = out.view(-1).view(torch.uint8)
+                flat.copy_(pinned[:nbytes], non_blocking=True)
+                if self.config.verify_crc and crc_expect is not None:
+                    crc_out = self._ext.crc32_async(flat)
+            done = self._copy_stream.record_event()
         done.synchronize()
-        if nbytes and self.config.verify_crc and crc_expect is not None:
-            crc = int(crc_t.item()) & 0xFFFFFFFF
+        if wire_fp8:
+            self._put_buf(self._dev_staging, dev_staging)
+        self._put_buf(self._pinned, pinned)
+        if self.config.verify_crc and crc_expect is not None:
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
             if crc != crc_expect:
                 raise ValueError(
                     f"GPU tensor CRC mismatch: expected {crc_expect:#x}, got {crc:#x}"
                 )
-        self._put_pinned(staging)
         return out
 
 

@@ -133,6 +133,12 @@ def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
             "nbytes": t.numel() * t.element_size(),
         }
         if gpu_plane is not None and t.device.type == "cuda":
+            if (
+                gpu_plane.config.wire_dtype == "fp8e4m3"
+                and t.dtype == torch.bfloat16
+            ):
+                man["wire"] = "fp8e4m3"
+                man["nbytes"] = t.numel()  # 1 byte/elt on the wire
             raw, crc = gpu_plane.pack_to_host(t)
             man["crc32"] = crc
         else:
@@ -173,7 +179,9 @@ def decode(
             and gpu_plane.config.place_on_gpu
         )
         if want_gpu:
-            t = gpu_plane.unpack_from_host(raw, dtype, man["shape"], man.get("crc32"))
+            t = gpu_plane.unpack_from_host(
+                raw, dtype, man["shape"], man.get("crc32"), man.get("wire")
+            )
         else:
             crc_expect = man.get("crc32")
             if crc_expect is not None and (gpu_plane is None or gpu_plane.config.verify_crc):
@@ -184,11 +192,22 @@ def decode(
                     raise ValueError(
                         f"tensor CRC mismatch: expected {crc_expect:#x}, got {crc:#x}"
                     )
-            t = torch.frombuffer(
-                bytearray(raw), dtype=torch.uint8
-            ).view(dtype).reshape(man["shape"]) if nbytes else torch.empty(
-                man["shape"], dtype=dtype
-            )
+            if man.get("wire") == "fp8e4m3" and nbytes:
+                # CPU fallback for fp8-compressed wire payloads.
+                t = (
+                    torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+                    .view(torch.float8_e4m3fn)
+                    .to(dtype)
+                    .reshape(man["shape"])
+                )
+            elif nbytes:
+                t = (
+                    torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+                    .view(dtype)
+                    .reshape(man["shape"])
+                )
+            else:
+                t = torch.empty(man["shape"], dtype=dtype)
         tensors.append(t)
 
     _decode_ctx.tensors = tensors

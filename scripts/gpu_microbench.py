@@ -1,0 +1,122 @@
+#!/usr/bin/env python3
+"""Data-plane kernel microbenchmarks on MI355X (run under gpurun).
+
+Measures: CRC32 kernel GB/s, fused pack+CRC, fp8 cast+CRC, FedAvg reduce,
+pinned D2H/H2D DMA, and the full pack_to_host/unpack_from_host pipeline.
+"""
+import json
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+from rayfed_amd.config import GpuDataPlaneConfig  # noqa: E402
+from rayfed_amd.ops import _hip_loader  # noqa: E402
+from rayfed_amd.ops.gpu_plane import GpuDataPlane  # noqa: E402
+
+ext = _hip_loader.load()
+
+
+def timeit(fn, reps=5, warm=2):
+    for _ in range(warm):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / reps
+
+
+def report(name, nbytes, secs, extra=None):
+    row = {"op": name, "MiB": nbytes >> 20, "GB/s": round(nbytes / secs / 1e9, 1),
+           "ms": round(secs * 1e3, 3)}
+    if extra:
+        row.update(extra)
+    print(json.dumps(row), flush=True)
+
+
+def main():
+    torch.cuda.set_device(0)
+    sizes = [1 << 24, 1 << 27, 1 << 30]  # 16 MiB, 128 MiB, 1 GiB
+
+    for n in sizes:
+        data = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        report("crc32", n, timeit(lambda: ext.crc32(data)))
+
+    for n in sizes:
+        src = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        dst = torch.empty(n, dtype=torch.uint8, device="cuda")
+        out = [None]
+
+        def run():
+            out[0] = ext.pack_crc_async(src, dst)
+
+        # read+write pass: report as 2x bytes moved
+        report("pack_crc(fused copy+crc)", 2 * n, timeit(run))
+
+    for n_elems in [1 << 24, 1 << 28]:  # bf16 elements
+        src = torch.randn(n_elems, device="cuda").to(torch.bfloat16)
+        dst = torch.empty(n_elems, dtype=torch.uint8, device="cuda")
+
+        def run():
+            ext.pack_fp8_async(src, dst)
+
+        report("pack_fp8(cast+crc)", 3 * n_elems, timeit(run),
+               {"elems": n_elems})
+
+    # FedAvg reduce: k inputs + 1 output of bf16
+    for k in [2, 4, 8]:
+        n = 1 << 28  # elements
+        ins = [torch.randn(n, device="cuda").to(torch.bfloat16) for _ in range(k)]
+        outt = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+        w = [1.0 / k] * k
+
+        def run():
+            ext.fedavg_reduce_(outt, ins, w)
+
+        moved = (k + 1) * n * 2
+        report(f"fedavg_reduce k={k} bf16", moved, timeit(run), {"elems": n})
+
+    # DMA: pinned D2H / H2D
+    n = 1 << 30
+    dev = torch.empty(n, dtype=torch.uint8, device="cuda")
+    pin = torch.empty(n, dtype=torch.uint8, pin_memory=True)
+    report("D2H pinned", n, timeit(lambda: pin.copy_(dev, non_blocking=False)))
+    report("H2D pinned", n, timeit(lambda: dev.copy_(pin, non_blocking=False)))
+
+    # Full plane pipeline
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    for gib in [0.25, 1.0]:
+        numel = int(gib * (1 << 30)) // 2
+        t = torch.randn(numel, device="cuda").to(torch.bfloat16)
+        nbytes = numel * 2
+
+        def pack():
+            plane.pack_to_host(t)
+
+        secs = timeit(pack, reps=3, warm=1)
+        report(f"plane.pack_to_host {gib} GiB bf16", nbytes, secs)
+
+        raw, crc = plane.pack_to_host(t)
+
+        def unpack():
+            plane.unpack_from_host(memoryview(raw), torch.bfloat16, [numel], crc)
+
+        report(f"plane.unpack_from_host {gib} GiB bf16", nbytes,
+               timeit(unpack, reps=3, warm=1))
+
+    plane8 = GpuDataPlane(GpuDataPlaneConfig(wire_dtype="fp8e4m3"))
+    numel = (1 << 29)  # 1 GiB bf16 -> 512 MiB wire
+    t = torch.randn(numel, device="cuda").to(torch.bfloat16)
+
+    def pack8():
+        plane8.pack_to_host(t)
+
+    report("plane.pack_to_host fp8-wire 1GiB bf16", numel * 2,
+           timeit(pack8, reps=3, warm=1))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,32 @@
+"""Build the in-tree HIP extension for gfx950.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built ``rayfed_amd/_hip*.so`` is git-ignored but travels with repo
+snapshots to GPU boxes (it must live in-tree, not in site-packages).
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils import cpp_extension  # noqa: E402
+
+setup(
+    name="rayfed_amd",
+    version="0.1.0",
+    packages=["rayfed_amd"],
+    ext_modules=[
+        cpp_extension.CUDAExtension(
+            name="rayfed_amd._hip",
+            sources=["csrc/rayfed_hip.hip"],
+            include_dirs=[os.path.abspath("csrc")],
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": cpp_extension.BuildExtension},
+)
