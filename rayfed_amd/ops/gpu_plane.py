@@ -79,10 +79,13 @@ class GpuDataPlane:
         del buf
 
     # -- send path ------------------------------------------------------------
-    def pack_to_host(self, t: "torch.Tensor") -> Tuple[bytes, Optional[int]]:
+    def pack_to_host(self, t: "torch.Tensor"):
         """Flatten ``t`` to raw wire bytes in pinned host memory.
 
-        Returns (bytes-like view over pinned staging, crc32|None).  The CRC
+        Returns (memoryview over pinned staging, crc32|None, release_fn).
+        Zero host-side copies: the view aliases the pooled pinned buffer and
+        the CALLER must invoke ``release_fn()`` once the bytes have left the
+        process (transport ack) to return the buffer to the pool.  The CRC
         kernel and the D2H DMA run on separate streams off the producing
         stream's event, overlapping each other.
         """
@@ -91,7 +94,7 @@ class GpuDataPlane:
             t = t.contiguous()
         nbytes = t.numel() * t.element_size()
         if nbytes == 0:
-            return b"", (0 if self.config.verify_crc else None)
+            return memoryview(b""), (0 if self.config.verify_crc else None), None
         flat = t.view(-1).view(torch.uint8)
 
         wire_fp8 = (
@@ -116,9 +119,8 @@ class GpuDataPlane:
             done.synchronize()
             self._put_buf(self._dev_staging, dev_staging)
             crc = int(crc_out[2].item()) & 0xFFFFFFFF if self.config.verify_crc else None
-            out = pinned[:wire_bytes].numpy().tobytes()
-            self._put_buf(self._pinned, pinned)
-            return out, crc
+            view = memoryview(pinned.numpy())[:wire_bytes]
+            return view, crc, lambda: self._put_buf(self._pinned, pinned)
 
         # Same-dtype wire: CRC pass and D2H DMA read `flat` concurrently.
         pinned = self._get_buf(self._pinned, nbytes, pin=True)
@@ -136,9 +138,8 @@ class GpuDataPlane:
         if self.config.verify_crc:
             crc_done.synchronize()
             crc = int(crc_out[2].item()) & 0xFFFFFFFF
-        out = pinned[:nbytes].numpy().tobytes()
-        self._put_buf(self._pinned, pinned)
-        return out, crc
+        view = memoryview(pinned.numpy())[:nbytes]
+        return view, crc, lambda: self._put_buf(self._pinned, pinned)
 
     # -- recv path ------------------------------------------------------------
     def unpack_from_host(

@@ -107,11 +107,13 @@ def _tensor_bytes_cpu(t: "torch.Tensor") -> bytes:
 def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
     """Serialize ``obj``; returns (header_extras, payload_parts).
 
-    ``payload_parts`` is a list of buffers to be concatenated on the wire
-    (skeleton pickle first, then each tensor's raw bytes).  When
-    ``gpu_plane`` is given (a :class:`rayfed_amd.ops.gpu_plane.GpuDataPlane`),
-    device tensors are packed + CRC'd by the HIP kernel and staged through
-    pinned memory asynchronously; otherwise a CPU copy is used.
+    ``payload_parts`` is a list of buffers to be written sequentially on the
+    wire (skeleton pickle first, then each tensor's raw bytes).  GPU-packed
+    parts are zero-copy views over pooled pinned staging buffers: the caller
+    MUST call :func:`release_parts` when the bytes have left the process
+    (after the transport ack), which returns the staging to the pool.
+    When ``gpu_plane`` is None or tensors are CPU-resident, parts are plain
+    host bytes and release is a no-op.
     """
     buf = io.BytesIO()
     pickler = _TensorExtractingPickler(buf)
@@ -120,6 +122,7 @@ def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
 
     manifests: List[Dict] = []
     parts: List[memoryview] = [memoryview(skeleton)]
+    releases: List = []
     for t in pickler.tensors:
         if torch is None:
             raise RuntimeError("torch payload without torch installed")
@@ -139,8 +142,10 @@ def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
             ):
                 man["wire"] = "fp8e4m3"
                 man["nbytes"] = t.numel()  # 1 byte/elt on the wire
-            raw, crc = gpu_plane.pack_to_host(t)
+            raw, crc, release = gpu_plane.pack_to_host(t)
             man["crc32"] = crc
+            if release is not None:
+                releases.append(release)
         else:
             raw = _tensor_bytes_cpu(t)
             if gpu_plane is not None and gpu_plane.config.verify_crc:
@@ -150,7 +155,18 @@ def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
         manifests.append(man)
         parts.append(memoryview(raw))
     extras = {"skel": len(skeleton), "tensors": manifests}
+    if releases:
+        extras["_releases"] = releases  # stripped before hitting the wire
     return extras, parts
+
+
+def release_parts(extras: Dict) -> None:
+    """Return pooled staging buffers referenced by encode() output."""
+    for rel in extras.pop("_releases", []):
+        try:
+            rel()
+        except Exception:  # noqa: BLE001 - pool return must never raise
+            pass
 
 
 def decode(

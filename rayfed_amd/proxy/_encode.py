@@ -2,16 +2,36 @@
 from __future__ import annotations
 
 import asyncio
-from typing import Optional
+from typing import List, Optional
 
 from rayfed_amd._private import serialization
 from rayfed_amd.exceptions import FedRemoteError
 from rayfed_amd.ops import tensor_codec
 from rayfed_amd.proxy.grpc import frames
 
-# Below this (estimated) payload size, encode inline on the I/O loop — the
-# executor hop costs more than the pickle.
-_INLINE_HINT = object()
+
+class EncodedRequest:
+    """A framed request as (prefix, payload parts): the TCP transport writes
+    parts sequentially (no join copy — GPU payload parts are views over
+    pinned staging); :meth:`release` returns pooled buffers after the ack."""
+
+    __slots__ = ("prefix", "parts", "_extras")
+
+    def __init__(self, prefix: bytes, parts: List, extras: Optional[dict] = None):
+        self.prefix = prefix
+        self.parts = parts
+        self._extras = extras
+
+    @property
+    def total_len(self) -> int:
+        return len(self.prefix) + sum(len(p) for p in self.parts)
+
+    def to_bytes(self) -> bytes:
+        return self.prefix + b"".join(bytes(p) for p in self.parts)
+
+    def release(self) -> None:
+        if self._extras is not None:
+            tensor_codec.release_parts(self._extras)
 
 
 async def encode_request(
@@ -21,7 +41,7 @@ async def encode_request(
     downstream_seq_id,
     gpu_plane=None,
     extra_header: Optional[dict] = None,
-) -> bytes:
+) -> EncodedRequest:
     header = {
         "job": job_name,
         "up": str(upstream_seq_id),
@@ -30,15 +50,16 @@ async def encode_request(
     if extra_header:
         header.update(extra_header)
     if isinstance(data, FedRemoteError):
-        return frames.encode_frame(
-            frames.KIND_ERROR, header, serialization.dumps(data)
+        payload = serialization.dumps(data)
+        return EncodedRequest(
+            frames.encode_frame_prefix(frames.KIND_ERROR, header), [payload]
         )
 
     # Cheap scalar/bytes payloads encode inline (no executor hop); anything
     # else — containers, tensors, user objects — encodes in the pool so a
     # multi-GiB pickle or a GPU pack never blocks the I/O loop.
-    if isinstance(data, (int, float, str, bytes, bool, type(None))) and (
-        not isinstance(data, (str, bytes)) or len(data) < 64 * 1024
+    if isinstance(data, (int, float, bool, type(None))) or (
+        isinstance(data, (str, bytes)) and len(data) < 64 * 1024
     ):
         extras, parts = tensor_codec.encode(data, gpu_plane)
     else:
@@ -47,8 +68,14 @@ async def encode_request(
             None, tensor_codec.encode, data, gpu_plane
         )
     if extras["tensors"]:
-        header.update(extras)
-        return frames.encode_frame(
-            frames.KIND_TENSOR, header, b"".join(bytes(p) for p in parts)
+        releases_extras = extras if "_releases" in extras else None
+        wire_header = {k: v for k, v in extras.items() if k != "_releases"}
+        header.update(wire_header)
+        return EncodedRequest(
+            frames.encode_frame_prefix(frames.KIND_TENSOR, header),
+            parts,
+            releases_extras,
         )
-    return frames.encode_frame(frames.KIND_PICKLE, header, parts[0])
+    return EncodedRequest(
+        frames.encode_frame_prefix(frames.KIND_PICKLE, header), [parts[0]]
+    )

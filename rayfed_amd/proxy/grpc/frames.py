@@ -55,6 +55,13 @@ def encode_frame(
     return _PREFIX.pack(MAGIC, VERSION, kind, 0, len(hdr)) + hdr + bytes(payload)
 
 
+def encode_frame_prefix(kind: int, header: Dict[str, Any]) -> bytes:
+    """Frame prefix only — the payload parts are written separately by the
+    transport (zero-copy streaming of pinned staging views)."""
+    hdr = msgpack.packb(header, use_bin_type=True)
+    return _PREFIX.pack(MAGIC, VERSION, kind, 0, len(hdr)) + hdr
+
+
 def decode_frame(data: bytes) -> Tuple[int, Dict[str, Any], memoryview]:
     """Return (kind, header, payload-view).  The payload is a zero-copy view
     into the request buffer."""

@@ -87,9 +87,9 @@ class GrpcSenderProxy(base_proxy.SenderProxy):
     # -- sending --------------------------------------------------------------
     async def send(self, dest_party, data, upstream_seq_id, downstream_seq_id):
         stub = self._get_stub(dest_party)
-        request = await self._encode_request(
-            data, upstream_seq_id, downstream_seq_id
-        )
+        req = await self._encode_request(data, upstream_seq_id, downstream_seq_id)
+        request = req.to_bytes()
+        req.release()  # joined into one protobuf-free bytes body
         timeout = 60.0
         if self._proxy_config is not None and self._proxy_config.timeout_in_ms:
             timeout = self._proxy_config.timeout_in_ms / 1000.0
@@ -196,7 +196,7 @@ class GrpcReceiverProxy(base_proxy.ReceiverProxy):
             kind, header, payload = frames.decode_frame(request)
         except ValueError as e:
             return frames.encode_response(400, f"bad frame: {e}")
-        code, result = self._mailbox.deliver(kind, header, bytes(payload))
+        code, result = self._mailbox.deliver(kind, header, payload)
         if code == 417:
             logger.warning("Rejected message: %s", result)
         return frames.encode_response(code, result)

@@ -20,10 +20,12 @@ Semantics preserved from the reference transport
   (``ca_cert``/``key``/``cert`` (+ ``target_name_override``)).
 
 Wire format (little-endian):
-  request :  u32 length ‖ frame (frames.py — header carries a request "id")
+  request :  u64 length ‖ u64 request-id ‖ frame (frames.py)
   response:  u32 length ‖ msgpack {id, code, result}
-Responses may arrive out of order; the "id" correlates them, so many sends
-pipeline on one connection.
+The u64 request length admits >4 GiB tensor pushes; responses may arrive out
+of order ("id" correlates them), so many sends pipeline on one connection.
+Large payload parts (views over pinned GPU staging) are written sequentially
+without any join copy.
 """
 from __future__ import annotations
 
@@ -120,16 +122,27 @@ class _Connection:
                 fut.set_exception(ConnectionError(str(exc)))
         self.pending.clear()
 
-    async def request(self, frame: bytes, timeout: float) -> dict:
+    async def request(self, prefix: bytes, parts, timeout: float) -> dict:
         req_id = next(self.ids)
         fut = asyncio.get_running_loop().create_future()
         self.pending[req_id] = fut
-        # The id rides ahead of the frame so the frame body stays opaque.
-        self.writer.write(
-            (len(frame) + 8).to_bytes(_LEN, "little")
-            + req_id.to_bytes(8, "little")
-            + frame
-        )
+        total = len(prefix) + sum(len(p) for p in parts)
+        head = (total + 8).to_bytes(8, "little") + req_id.to_bytes(8, "little")
+        _CHUNK = 8 << 20  # bound transport buffering for multi-GiB parts
+        if total <= 65536:
+            # Small request: one write, one TCP segment.
+            self.writer.write(head + prefix + b"".join(bytes(p) for p in parts))
+        else:
+            self.writer.write(head)
+            self.writer.write(prefix)
+            for p in parts:
+                if len(p) <= _CHUNK:
+                    self.writer.write(p)
+                else:
+                    mv = memoryview(p)
+                    for off in range(0, len(mv), _CHUNK):
+                        self.writer.write(mv[off : off + _CHUNK])
+                        await self.writer.drain()
         await self.writer.drain()
         return await asyncio.wait_for(fut, timeout=timeout)
 
@@ -183,6 +196,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
                 int(port),
                 ssl=self._ssl,
                 server_hostname=self._server_hostname if self._ssl else None,
+                limit=16 << 20,
             )
             writer.transport.set_write_buffer_limits(high=1 << 26)
             conn = _Connection(reader, writer)
@@ -190,9 +204,15 @@ class TcpSenderProxy(base_proxy.SenderProxy):
             return conn
 
     async def send(self, dest_party, data, upstream_seq_id, downstream_seq_id):
-        frame = await encode_request(
+        req = await encode_request(
             self._job_name, data, upstream_seq_id, downstream_seq_id, self.gpu_plane
         )
+        try:
+            return await self._send_framed(dest_party, req)
+        finally:
+            req.release()  # pinned staging back to the pool after the ack
+
+    async def _send_framed(self, dest_party, req):
         deadline = time.monotonic() + self._timeout_s
         backoff = self._retry.initial_backoff
         attempt = 0
@@ -203,7 +223,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
                 remaining = deadline - time.monotonic()
                 if remaining <= 0:
                     raise TimeoutError(f"send to {dest_party} deadline exceeded")
-                resp = await conn.request(frame, timeout=remaining)
+                resp = await conn.request(req.prefix, req.parts, timeout=remaining)
                 code = resp.get("code", 500)
                 if 400 <= code < 500:
                     raise RuntimeError(
@@ -270,7 +290,7 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
         try:
             self._server = await asyncio.start_server(
                 self._handle_conn, host=None, port=port, ssl=ssl_ctx,
-                reuse_address=False,
+                reuse_address=False, limit=16 << 20,
             )
         except OSError as e:
             raise AssertionError(
@@ -287,13 +307,15 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
     async def _handle_conn(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
         try:
             while True:
-                hdr = await reader.readexactly(_LEN)
+                hdr = await reader.readexactly(8)
                 n = int.from_bytes(hdr, "little")
                 body = await reader.readexactly(n)
                 req_id = int.from_bytes(body[:8], "little")
                 try:
+                    # Zero-copy: the mailbox holds a view into the immutable
+                    # request body until the reader consumes it.
                     kind, header, payload = frames.decode_frame(body[8:])
-                    code, result = self._mailbox.deliver(kind, header, bytes(payload))
+                    code, result = self._mailbox.deliver(kind, header, payload)
                 except ValueError as e:
                     code, result = 400, f"bad frame: {e}"
                 if code == 417:

@@ -94,12 +94,14 @@ def main():
         nbytes = numel * 2
 
         def pack():
-            plane.pack_to_host(t)
+            _, _, rel = plane.pack_to_host(t)
+            if rel:
+                rel()
 
         secs = timeit(pack, reps=3, warm=1)
         report(f"plane.pack_to_host {gib} GiB bf16", nbytes, secs)
 
-        raw, crc = plane.pack_to_host(t)
+        raw, crc, _rel = plane.pack_to_host(t)
 
         def unpack():
             plane.unpack_from_host(memoryview(raw), torch.bfloat16, [numel], crc)
@@ -112,7 +114,9 @@ def main():
     t = torch.randn(numel, device="cuda").to(torch.bfloat16)
 
     def pack8():
-        plane8.pack_to_host(t)
+        _, _, rel = plane8.pack_to_host(t)
+        if rel:
+            rel()
 
     report("plane.pack_to_host fp8-wire 1GiB bf16", numel * 2,
            timeit(pack8, reps=3, warm=1))

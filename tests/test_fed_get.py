@@ -142,10 +142,18 @@ def _driver_ping(party, addresses):
     )
 
     @fed.remote
-    def f():
-        return "pong"
+    def f(v):
+        return v
 
-    assert fed.get(f.party("alice").remote()) == "pong"
+    @fed.remote
+    def join(a, b):
+        return a + b
+
+    # Cross-party join so neither driver can finish (and tear down its
+    # receiver) before the other has passed its own ping barrier.
+    a = f.party("alice").remote("po")
+    b = f.party("bob").remote("ng")
+    assert fed.get(join.party("bob").remote(a, b)) == "pong"
     fed.shutdown()
 
 

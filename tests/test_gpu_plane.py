@@ -118,10 +118,12 @@ def test_masked_add_matches_reference(ext):
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32, torch.int64])
 def test_plane_roundtrip(plane, dtype):
     t = (torch.arange(300_000, device="cuda") % 97).to(dtype)
-    raw, crc = plane.pack_to_host(t)
+    raw, crc, release = plane.pack_to_host(t)
     assert crc is not None
     assert crc == (zlib.crc32(raw) & 0xFFFFFFFF)
     back = plane.unpack_from_host(memoryview(raw), dtype, [300_000], crc)
+    if release:
+        release()
     assert back.device.type == "cuda"
     assert torch.equal(back, t)
 
@@ -129,8 +131,10 @@ def test_plane_roundtrip(plane, dtype):
 @needs_gpu
 def test_plane_crc_tamper_detected(plane):
     t = torch.randn(100_000, device="cuda")
-    raw, crc = plane.pack_to_host(t)
+    raw, crc, release = plane.pack_to_host(t)
     bad = bytearray(raw)
+    if release:
+        release()
     bad[1234] ^= 0xFF
     with pytest.raises(ValueError, match="CRC"):
         plane.unpack_from_host(memoryview(bytes(bad)), torch.float32, [100_000], crc)
@@ -143,11 +147,13 @@ def test_plane_fp8_wire(plane):
 
     p8 = GpuDataPlane(GpuDataPlaneConfig(wire_dtype="fp8e4m3"))
     t = (torch.randn(65536, device="cuda") * 2).to(torch.bfloat16)
-    raw, crc = p8.pack_to_host(t)
+    raw, crc, release = p8.pack_to_host(t)
     assert len(raw) == t.numel()  # 1 byte per element on the wire
     back = p8.unpack_from_host(
         memoryview(raw), torch.bfloat16, [65536], crc, wire_dtype="fp8e4m3"
     )
+    if release:
+        release()
     ref = t.to(torch.float8_e4m3fn).to(torch.bfloat16)
     close = torch.isclose(back.float(), ref.float(), atol=0.0, rtol=0.0)
     assert close.float().mean().item() > 0.999
@@ -163,6 +169,7 @@ def test_codec_end_to_end_gpu(plane):
     }
     extras, parts = tensor_codec.encode(obj, plane)
     payload = b"".join(bytes(p) for p in parts)
+    tensor_codec.release_parts(extras)
     out = tensor_codec.decode(extras, memoryview(payload), plane, None)
     assert out["w"].device.type == "cuda"
     assert torch.equal(out["w"], obj["w"])

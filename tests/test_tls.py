@@ -58,7 +58,7 @@ def _driver_plaintext_to_tls_fails(party, addresses, tls_config):
     try:
         with socket.create_connection((host, int(port)), timeout=5) as s:
             body = (1).to_bytes(8, "little") + raw
-            s.sendall(len(body).to_bytes(4, "little") + body)
+            s.sendall(len(body).to_bytes(8, "little") + body)
             s.settimeout(5)
             data = s.recv(4)
             # A TLS server must not complete our plaintext "request"; any

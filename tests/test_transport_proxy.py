@@ -92,7 +92,7 @@ def test_wrong_job_name_rejected_with_417(party_env):
         host, port = addrs["alice"].rsplit(":", 1)
         with socket.create_connection((host, int(port)), timeout=10) as s:
             body = (1).to_bytes(8, "little") + raw
-            s.sendall(len(body).to_bytes(4, "little") + body)
+            s.sendall(len(body).to_bytes(8, "little") + body)
             n = int.from_bytes(_read_n(s, 4), "little")
             resp = msgpack.unpackb(_read_n(s, n), raw=False)
     assert resp["code"] == 417
