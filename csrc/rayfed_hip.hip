@@ -596,6 +596,17 @@ int64_t crc32_combine_py(int64_t crc1, int64_t crc2, int64_t len2) {
                                             (uint64_t)len2);
 }
 
+// Pin an existing host mapping (e.g. a /dev/shm segment) for true-DMA
+// D2H/H2D — the same-host shm lane registers pooled segments once.
+void host_register(int64_t ptr, int64_t size) {
+  HIP_CHECK(hipHostRegister(reinterpret_cast<void*>(ptr), (size_t)size,
+                            hipHostRegisterDefault));
+}
+
+void host_unregister(int64_t ptr) {
+  HIP_CHECK(hipHostUnregister(reinterpret_cast<void*>(ptr)));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -613,4 +624,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "out = sum_k w_k * in_k (bf16/f16/f32, fp32 accumulation)");
   m.def("masked_add_", &masked_add_, "dst += mask ? src : 0");
   m.def("crc32_combine", &crc32_combine_py, "zlib-style CRC combine");
+  m.def("host_register", &host_register, "hipHostRegister an existing mapping");
+  m.def("host_unregister", &host_unregister, "hipHostUnregister");
 }

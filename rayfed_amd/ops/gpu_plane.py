@@ -141,6 +141,66 @@ class GpuDataPlane:
         view = memoryview(pinned.numpy())[:nbytes]
         return view, crc, lambda: self._put_buf(self._pinned, pinned)
 
+    def pack_to_shm(self, t: "torch.Tensor"):
+        """Like :meth:`pack_to_host` but the destination is a pooled
+        /dev/shm segment (hipHostRegister-ed once) so a same-host peer can
+        H2D straight out of it.  Returns (segment, crc|None, release_fn)."""
+        from rayfed_amd.ops import shm_pool
+
+        t = t.detach()
+        if not t.is_contiguous():
+            t = t.contiguous()
+        pool = shm_pool.get_send_pool()
+        wire_fp8 = (
+            self.config.wire_dtype == "fp8e4m3" and t.dtype == torch.bfloat16
+        )
+        produced = torch.cuda.current_stream(self.device).record_event()
+        crc_out = None
+        if wire_fp8:
+            wire_bytes = t.numel()
+            seg = pool.acquire(wire_bytes)
+            dev_staging = self._get_buf(self._dev_staging, wire_bytes, pin=False)
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+                crc_out = self._ext.pack_fp8_async(
+                    t.view(-1), dev_staging[:wire_bytes]
+                )
+                packed = self._crc_stream.record_event()
+            with torch.cuda.stream(self._copy_stream):
+                self._copy_stream.wait_event(packed)
+                seg.torch_view[:wire_bytes].copy_(
+                    dev_staging[:wire_bytes], non_blocking=seg.registered
+                )
+                done = self._copy_stream.record_event()
+            done.synchronize()
+            self._put_buf(self._dev_staging, dev_staging)
+            crc = (
+                int(crc_out[2].item()) & 0xFFFFFFFF
+                if self.config.verify_crc
+                else None
+            )
+            return seg, crc, lambda: pool.release(seg)
+
+        nbytes = t.numel() * t.element_size()
+        flat = t.view(-1).view(torch.uint8)
+        seg = pool.acquire(nbytes)
+        if self.config.verify_crc and nbytes:
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+                crc_out = self._ext.crc32_async(flat)
+                crc_done = self._crc_stream.record_event()
+        with torch.cuda.stream(self._copy_stream):
+            self._copy_stream.wait_event(produced)
+            if nbytes:
+                seg.torch_view[:nbytes].copy_(flat, non_blocking=seg.registered)
+            copy_done = self._copy_stream.record_event()
+        copy_done.synchronize()
+        crc = None
+        if self.config.verify_crc and nbytes:
+            crc_done.synchronize()
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
+        return seg, crc, lambda: pool.release(seg)
+
     # -- recv path ------------------------------------------------------------
     def unpack_from_host(
         self,
@@ -149,11 +209,18 @@ class GpuDataPlane:
         shape: List[int],
         crc_expect: Optional[int],
         wire_dtype: Optional[str] = None,
+        src_tensor: Optional["torch.Tensor"] = None,
     ) -> "torch.Tensor":
         nbytes = len(raw)
         out = torch.empty(shape, dtype=dtype, device=self.device)
         if nbytes == 0:
             return out
+        if src_tensor is not None:
+            # Registered shm segment: H2D DMA straight from the mapping —
+            # no staging copy.
+            return self._unpack_from_pinned(
+                src_tensor, out, nbytes, crc_expect, wire_dtype
+            )
         pinned = self._get_buf(self._pinned, nbytes, pin=True)
         pinned[:nbytes].numpy()[:] = memoryview(raw).cast("B")
 
@@ -176,6 +243,40 @@ class GpuDataPlane:
             self._put_buf(self._dev_staging, dev_staging)
         self._put_buf(self._pinned, pinned)
         if self.config.verify_crc and crc_expect is not None:
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
+            if crc != crc_expect:
+                raise ValueError(
+                    f"GPU tensor CRC mismatch: expected {crc_expect:#x}, got {crc:#x}"
+                )
+        return out
+
+    def _unpack_from_pinned(
+        self,
+        src: "torch.Tensor",
+        out: "torch.Tensor",
+        nbytes: int,
+        crc_expect: Optional[int],
+        wire_dtype: Optional[str],
+    ) -> "torch.Tensor":
+        wire_fp8 = wire_dtype == "fp8e4m3"
+        crc_out = None
+        with torch.cuda.stream(self._copy_stream):
+            if wire_fp8:
+                dev_staging = self._get_buf(self._dev_staging, nbytes, pin=False)
+                dev_staging[:nbytes].copy_(src[:nbytes], non_blocking=True)
+                if self.config.verify_crc and crc_expect is not None:
+                    crc_out = self._ext.crc32_async(dev_staging[:nbytes])
+                self._ext.unpack_fp8_async(dev_staging[:nbytes], out.view(-1))
+            else:
+                flat = out.view(-1).view(torch.uint8)
+                flat.copy_(src[:nbytes], non_blocking=True)
+                if self.config.verify_crc and crc_expect is not None:
+                    crc_out = self._ext.crc32_async(flat)
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        if wire_fp8:
+            self._put_buf(self._dev_staging, dev_staging)
+        if crc_out is not None:
             crc = int(crc_out[2].item()) & 0xFFFFFFFF
             if crc != crc_expect:
                 raise ValueError(

@@ -92,19 +92,21 @@ def contains_tensors(tensors: List) -> bool:
     return len(tensors) > 0
 
 
-def _tensor_bytes_cpu(t: "torch.Tensor") -> bytes:
-    """Contiguous raw bytes of a tensor via an untyped-storage view (one copy
-    off-device max; no pickle overhead)."""
+def _tensor_bytes_cpu(t: "torch.Tensor"):
+    """Raw bytes of a tensor as a zero-copy numpy view (the view keeps the
+    backing memory alive through the buffer protocol); GPU tensors cost one
+    D2H copy, CPU tensors none."""
     t = t.detach()
     if t.device.type != "cpu":
         t = t.to("cpu", non_blocking=False)
     if not t.is_contiguous():
         t = t.contiguous()
-    n = t.numel() * t.element_size()
-    return t.view(-1).view(torch.uint8).numpy().tobytes() if t.numel() else b""
+    if not t.numel():
+        return b""
+    return t.view(-1).view(torch.uint8).numpy()
 
 
-def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
+def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memoryview]]:
     """Serialize ``obj``; returns (header_extras, payload_parts).
 
     ``payload_parts`` is a list of buffers to be written sequentially on the
@@ -135,6 +137,43 @@ def encode(obj: Any, gpu_plane=None) -> Tuple[Dict, List[memoryview]]:
             "device": t.device.type,
             "nbytes": t.numel() * t.element_size(),
         }
+        nbytes = man["nbytes"]
+        use_shm = False
+        if shm:
+            from rayfed_amd.ops import shm_pool
+
+            use_shm = nbytes >= shm_pool.SHM_MIN_BYTES
+        if use_shm:
+            # Same-host lane: bytes land in a pooled /dev/shm segment; only
+            # the (segment, offset) reference rides the socket.  The segment
+            # is recycled when the receiver's ack releases it.
+            from rayfed_amd.ops import shm_pool
+
+            if gpu_plane is not None and t.device.type == "cuda":
+                if (
+                    gpu_plane.config.wire_dtype == "fp8e4m3"
+                    and t.dtype == torch.bfloat16
+                ):
+                    man["wire"] = "fp8e4m3"
+                    man["nbytes"] = t.numel()
+                seg, crc, release = gpu_plane.pack_to_shm(t)
+                man["crc32"] = crc
+            else:
+                pool = shm_pool.get_send_pool()
+                seg = pool.acquire(nbytes)
+                raw_cpu = _tensor_bytes_cpu(t)
+                seg.array[: len(raw_cpu)] = memoryview(raw_cpu)
+                man["nbytes"] = len(raw_cpu)
+                if gpu_plane is not None and gpu_plane.config.verify_crc:
+                    import zlib
+
+                    man["crc32"] = zlib.crc32(raw_cpu) & 0xFFFFFFFF
+                release = lambda s=seg, p=pool: p.release(s)  # noqa: E731
+            man["shm"] = seg.name
+            man["shm_off"] = 0
+            releases.append(release)
+            manifests.append(man)
+            continue  # no payload part for shm-borne tensors
         if gpu_plane is not None and t.device.type == "cuda":
             if (
                 gpu_plane.config.wire_dtype == "fp8e4m3"
@@ -184,8 +223,14 @@ def decode(
     tensors: List[Any] = []
     for man in extras["tensors"]:
         nbytes = man["nbytes"]
-        raw = payload[off : off + nbytes]
-        off += nbytes
+        if "shm" in man:
+            from rayfed_amd.ops import shm_pool
+
+            seg = shm_pool.attach(man["shm"])
+            raw = seg.view(man["shm_off"], nbytes)
+        else:
+            raw = payload[off : off + nbytes]
+            off += nbytes
         if torch is None:
             raise RuntimeError("torch payload without torch installed")
         dtype = _STR_TO_DTYPE[man["dtype"]]
@@ -195,8 +240,17 @@ def decode(
             and gpu_plane.config.place_on_gpu
         )
         if want_gpu:
+            src_t = None
+            if "shm" in man:
+                from rayfed_amd.ops import shm_pool
+
+                seg = shm_pool.attach(man["shm"])
+                if seg.registered and seg.torch_view is not None:
+                    o = man["shm_off"]
+                    src_t = seg.torch_view[o : o + nbytes]
             t = gpu_plane.unpack_from_host(
-                raw, dtype, man["shape"], man.get("crc32"), man.get("wire")
+                raw, dtype, man["shape"], man.get("crc32"), man.get("wire"),
+                src_tensor=src_t,
             )
         else:
             crc_expect = man.get("crc32")

@@ -41,6 +41,7 @@ async def encode_request(
     downstream_seq_id,
     gpu_plane=None,
     extra_header: Optional[dict] = None,
+    shm: bool = False,
 ) -> EncodedRequest:
     header = {
         "job": job_name,
@@ -61,11 +62,11 @@ async def encode_request(
     if isinstance(data, (int, float, bool, type(None))) or (
         isinstance(data, (str, bytes)) and len(data) < 64 * 1024
     ):
-        extras, parts = tensor_codec.encode(data, gpu_plane)
+        extras, parts = tensor_codec.encode(data, gpu_plane, shm)
     else:
         loop = asyncio.get_running_loop()
         extras, parts = await loop.run_in_executor(
-            None, tensor_codec.encode, data, gpu_plane
+            None, tensor_codec.encode, data, gpu_plane, shm
         )
     if extras["tensors"]:
         releases_extras = extras if "_releases" in extras else None

@@ -27,15 +27,18 @@ class Mailbox:
         self.received_op_count = 0
         self.gpu_plane = None
 
-    def deliver(self, kind: int, header: dict, payload: bytes) -> Tuple[int, str]:
-        """Called by the transport on message arrival (on the I/O loop).
-        Returns (code, result) for the ack."""
+    _KIND_OBJ = -1  # already-decoded object (shm lane eager decode)
+
+    def check_job(self, header: dict) -> Optional[Tuple[int, str]]:
         job_name = header.get("job", "")
         if job_name != self._job_name:
             return (
                 417,
                 f"JobName mis-match: expected {self._job_name!r}, got {job_name!r}",
             )
+        return None
+
+    def _park(self, header: dict, item) -> None:
         key = (header["up"], header["down"])
         self.received_op_count += 1
         fut = self._slots.get(key)
@@ -43,7 +46,24 @@ class Mailbox:
             fut = asyncio.get_running_loop().create_future()
             self._slots[key] = fut
         if not fut.done():
-            fut.set_result((kind, header, payload))
+            fut.set_result(item)
+
+    def deliver(self, kind: int, header: dict, payload: bytes) -> Tuple[int, str]:
+        """Called by the transport on message arrival (on the I/O loop).
+        Returns (code, result) for the ack."""
+        bad = self.check_job(header)
+        if bad is not None:
+            return bad
+        self._park(header, (kind, header, payload))
+        return 200, "OK"
+
+    def deliver_obj(self, header: dict, obj) -> Tuple[int, str]:
+        """Deliver an already-decoded object (the shm lane decodes before
+        acking so the sender can recycle its segment on ack)."""
+        bad = self.check_job(header)
+        if bad is not None:
+            return bad
+        self._park(header, (self._KIND_OBJ, header, obj))
         return 200, "OK"
 
     async def get_data(self, upstream_seq_id, curr_seq_id):
@@ -55,6 +75,10 @@ class Mailbox:
         kind, header, payload = await fut
         self._slots.pop(key, None)
         loop = asyncio.get_running_loop()
+        if kind == self._KIND_OBJ:
+            if isinstance(payload, BaseException):
+                raise payload
+            return payload
         if kind == frames.KIND_ERROR:
             raise serialization.loads(payload, self._allowed_list)
         if kind == frames.KIND_TENSOR:

@@ -165,6 +165,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
         super().__init__(addresses, party, job_name, tls_config, proxy_config)
         self._conns: Dict[str, _Connection] = {}
         self._conn_locks: Dict[str, asyncio.Lock] = {}
+        self._same_host_cache: Dict[str, bool] = {}
         self.gpu_plane = None
         self._retry = _RetryPolicy(
             getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
@@ -203,9 +204,26 @@ class TcpSenderProxy(base_proxy.SenderProxy):
             self._conns[dest_party] = conn
             return conn
 
+    def _same_host(self, dest_party: str) -> bool:
+        """The shm lane applies when the peer shares this node (loopback or
+        local hostname) — the BASELINE topologies co-locate all parties."""
+        cached = self._same_host_cache.get(dest_party)
+        if cached is not None:
+            return cached
+        import socket
+
+        host = self._addresses.get(dest_party, "").rsplit(":", 1)[0]
+        same = host in ("127.0.0.1", "localhost", "::1", socket.gethostname())
+        self._same_host_cache[dest_party] = same
+        return same
+
     async def send(self, dest_party, data, upstream_seq_id, downstream_seq_id):
+        from rayfed_amd.ops import shm_pool
+
+        use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
         req = await encode_request(
-            self._job_name, data, upstream_seq_id, downstream_seq_id, self.gpu_plane
+            self._job_name, data, upstream_seq_id, downstream_seq_id,
+            self.gpu_plane, shm=use_shm,
         )
         try:
             return await self._send_framed(dest_party, req)
@@ -315,7 +333,16 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
                     # Zero-copy: the mailbox holds a view into the immutable
                     # request body until the reader consumes it.
                     kind, header, payload = frames.decode_frame(body[8:])
-                    code, result = self._mailbox.deliver(kind, header, payload)
+                    if kind == frames.KIND_TENSOR and any(
+                        "shm" in m for m in header.get("tensors", ())
+                    ):
+                        # shm lane: consume (H2D + CRC) BEFORE acking — the
+                        # ack licenses the sender to recycle its segment.
+                        code, result = await self._consume_shm_frame(
+                            header, payload
+                        )
+                    else:
+                        code, result = self._mailbox.deliver(kind, header, payload)
                 except ValueError as e:
                     code, result = 400, f"bad frame: {e}"
                 if code == 417:
@@ -334,6 +361,27 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
                 writer.close()
             except Exception:  # noqa: BLE001
                 pass
+
+    async def _consume_shm_frame(self, header, payload):
+        bad = self._mailbox.check_job(header)
+        if bad is not None:
+            return bad
+        from rayfed_amd.ops import tensor_codec
+
+        loop = asyncio.get_running_loop()
+        try:
+            obj = await loop.run_in_executor(
+                None,
+                tensor_codec.decode,
+                {k: header[k] for k in ("skel", "tensors")},
+                memoryview(payload),
+                self.gpu_plane,
+                self._mailbox._allowed_list,
+            )
+        except Exception as e:  # noqa: BLE001 — CRC mismatch, attach failure…
+            logger.warning("shm frame consume failed: %r", e)
+            return 500, f"shm consume failed: {e!r}"
+        return self._mailbox.deliver_obj(header, obj)
 
     async def get_data(self, src_party, upstream_seq_id, curr_seq_id):
         return await self._mailbox.get_data(upstream_seq_id, curr_seq_id)
