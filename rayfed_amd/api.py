@@ -122,24 +122,35 @@ def init(
         if cross_silo_comm_config.use_global_proxy is not None
         else True
     )
-    receiver = barriers.start_receiver_proxy(
-        addresses,
-        party,
-        job_name=job_name,
-        tls_config=tls_config,
-        proxy_cls=receiver_proxy_cls,
-        proxy_config=cross_silo_comm_config,
-        use_global_proxy=use_global_proxy,
-    )
-    sender = barriers.start_sender_proxy(
-        addresses,
-        party,
-        job_name=job_name,
-        tls_config=tls_config,
-        proxy_cls=sender_proxy_cls,
-        proxy_config=cross_silo_comm_config,
-        use_global_proxy=use_global_proxy,
-    )
+    if receiver_sender_proxy_cls is not None:
+        sender, receiver = barriers.start_sender_receiver_proxy(
+            addresses,
+            party,
+            job_name=job_name,
+            tls_config=tls_config,
+            proxy_cls=receiver_sender_proxy_cls,
+            proxy_config=cross_silo_comm_config,
+            use_global_proxy=use_global_proxy,
+        )
+    else:
+        receiver = barriers.start_receiver_proxy(
+            addresses,
+            party,
+            job_name=job_name,
+            tls_config=tls_config,
+            proxy_cls=receiver_proxy_cls,
+            proxy_config=cross_silo_comm_config,
+            use_global_proxy=use_global_proxy,
+        )
+        sender = barriers.start_sender_proxy(
+            addresses,
+            party,
+            job_name=job_name,
+            tls_config=tls_config,
+            proxy_cls=sender_proxy_cls,
+            proxy_config=cross_silo_comm_config,
+            use_global_proxy=use_global_proxy,
+        )
 
     # Attach the GPU data plane when a HIP device is visible: tensors then
     # ride the pack/CRC/pinned-staging path instead of pickle.

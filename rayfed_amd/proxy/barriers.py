@@ -257,6 +257,41 @@ def start_sender_proxy(
     return service
 
 
+def start_sender_receiver_proxy(
+    addresses: Dict,
+    party: str,
+    job_name: str,
+    tls_config: Optional[Dict] = None,
+    proxy_cls=None,
+    proxy_config: Optional[CrossSiloMessageConfig] = None,
+    ready_timeout_second: int = 60,
+    use_global_proxy: bool = True,
+):
+    """Combined single-service variant (parity: reference barriers.py:415-459).
+    Both module-level send() and recv() route through one proxy object."""
+    global _sender_service, _receiver_service
+    if proxy_cls is None:
+        from rayfed_amd.proxy.tcp.combined import TcpSenderReceiverProxy
+
+        proxy_cls = TcpSenderReceiverProxy
+    io = _get_io_loop()
+    proxy = proxy_cls(
+        addresses, addresses[party], party, job_name, tls_config, proxy_config
+    )
+    recv_service = ReceiverProxyService(proxy, io)
+    recv_service.start(ready_timeout_second=ready_timeout_second)
+    send_service = SenderProxyService(proxy, io)
+    _sender_service = send_service
+    _receiver_service = recv_service
+    name = (
+        _SENDER_RECEIVER_PROXY_NAME
+        if use_global_proxy
+        else f"{_SENDER_RECEIVER_PROXY_NAME}-{job_name}"
+    )
+    _service_registry[name] = (send_service, recv_service)
+    return send_service, recv_service
+
+
 def _cleanup_proxies():
     """Stop proxies and the I/O loop (called from fed.shutdown)."""
     global _sender_service, _receiver_service, _io_loop
