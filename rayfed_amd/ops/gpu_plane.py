@@ -141,6 +141,144 @@ class GpuDataPlane:
         view = memoryview(pinned.numpy())[:nbytes]
         return view, crc, lambda: self._put_buf(self._pinned, pinned)
 
+    # Chunk-pipelined shm pushes: the first SHM_HDR bytes of the segment are
+    # a progress page — u64 chunks-ready counter at offset 0, then one u32
+    # CRC per chunk at offset 8 — written by the sender as each chunk's D2H
+    # completes, so the receiver's H2D overlaps the sender's D2H.
+    SHM_HDR = 4096
+
+    def pack_to_shm_chunked(self, t: "torch.Tensor"):
+        """Start a chunked D2H into a pooled shm segment and return
+        IMMEDIATELY with (segment, manifest_fields, release_fn); a worker
+        thread advances the progress counter as chunk DMAs complete.
+        Only for plain-wire contiguous tensors above ~2 chunks."""
+        import struct
+
+        from rayfed_amd.ops import shm_pool
+
+        nbytes = t.numel() * t.element_size()
+        chunk = self.config.chunk_bytes
+        pool = shm_pool.get_send_pool()
+        seg = pool.acquire(self.SHM_HDR + nbytes)
+        n_chunks = (nbytes + chunk - 1) // chunk
+        seg.array[: self.SHM_HDR].fill(0)
+        flat = t.view(-1).view(torch.uint8)
+        produced = torch.cuda.current_stream(self.device).record_event()
+
+        def _drive():
+            try:
+                torch.cuda.set_device(self.device)
+                events = []
+                crc_outs = []
+                with torch.cuda.stream(self._copy_stream):
+                    self._copy_stream.wait_event(produced)
+                if self.config.verify_crc:
+                    with torch.cuda.stream(self._crc_stream):
+                        self._crc_stream.wait_event(produced)
+                for i in range(n_chunks):
+                    lo = i * chunk
+                    hi = min(lo + chunk, nbytes)
+                    with torch.cuda.stream(self._copy_stream):
+                        seg.torch_view[
+                            self.SHM_HDR + lo : self.SHM_HDR + hi
+                        ].copy_(flat[lo:hi], non_blocking=seg.registered)
+                        ev = torch.cuda.Event()
+                        ev.record(self._copy_stream)
+                    crc_out = None
+                    if self.config.verify_crc:
+                        with torch.cuda.stream(self._crc_stream):
+                            crc_out = self._ext.crc32_async(flat[lo:hi])
+                            cev = torch.cuda.Event()
+                            cev.record(self._crc_stream)
+                    events.append((ev, cev if self.config.verify_crc else None))
+                    crc_outs.append(crc_out)
+                for i, (ev, cev) in enumerate(events):
+                    ev.synchronize()
+                    if cev is not None:
+                        cev.synchronize()
+                        crc = int(crc_outs[i][2].item()) & 0xFFFFFFFF
+                        struct.pack_into("<I", seg.array, 8 + 4 * i, crc)
+                    # Publish chunk i (x86: aligned 8-byte store is atomic).
+                    struct.pack_into("<Q", seg.array, 0, i + 1)
+            except Exception:  # noqa: BLE001
+                logger.exception("chunked shm pack failed")
+                struct.pack_into("<q", seg.array, 0, -1)  # poison
+
+        driver = threading.Thread(target=_drive, daemon=True, name="shm-pack")
+        driver.start()
+
+        man_fields = {
+            "chunked": chunk,
+            "hdr": self.SHM_HDR,
+            "crc_per_chunk": bool(self.config.verify_crc),
+        }
+
+        def release():
+            driver.join(timeout=60)
+            pool.release(seg)
+
+        return seg, man_fields, release
+
+    def unpack_from_shm_chunked(self, seg_name: str, man, dtype, shape):
+        """Receiver side of the chunk pipeline: H2D each chunk as soon as
+        the sender publishes it, CRC-verify on device, overlap everything on
+        the copy stream."""
+        import struct
+        import time as _time
+
+        from rayfed_amd.ops import shm_pool
+
+        seg = shm_pool.attach(seg_name)
+        nbytes = man["nbytes"]
+        chunk = man["chunked"]
+        hdr = man["hdr"]
+        n_chunks = (nbytes + chunk - 1) // chunk
+        out = torch.empty(shape, dtype=dtype, device=self.device)
+        flat = out.view(-1).view(torch.uint8)
+        crc_outs = []
+        deadline = _time.monotonic() + 600
+        with torch.cuda.stream(self._copy_stream):
+            for i in range(n_chunks):
+                while True:
+                    ready = struct.unpack_from("<q", seg.array, 0)[0]
+                    if ready > i:
+                        break
+                    if ready < 0:
+                        raise RuntimeError("peer aborted chunked shm push")
+                    if _time.monotonic() > deadline:
+                        raise TimeoutError("chunked shm push stalled")
+                    _time.sleep(0.0002)
+                lo = i * chunk
+                hi = min(lo + chunk, nbytes)
+                src = (
+                    seg.torch_view[hdr + lo : hdr + hi]
+                    if seg.registered and seg.torch_view is not None
+                    else None
+                )
+                if src is not None:
+                    flat[lo:hi].copy_(src, non_blocking=True)
+                else:  # unregistered fallback: sync copy through a staging buf
+                    flat[lo:hi].copy_(
+                        torch.frombuffer(
+                            bytearray(seg.view(hdr + lo, hi - lo)),
+                            dtype=torch.uint8,
+                        )
+                    )
+                if man.get("crc_per_chunk"):
+                    crc_outs.append(self._ext.crc32_async(flat[lo:hi]))
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        if man.get("crc_per_chunk"):
+            for i, crc_out in enumerate(crc_outs):
+                got = int(crc_out[2].item()) & 0xFFFFFFFF
+                want = struct.unpack_from("<I", seg.array, 8 + 4 * i)[0]
+                if got != want:
+                    raise ValueError(
+                        f"GPU tensor CRC mismatch on chunk {i}: "
+                        f"expected {want:#x}, got {got:#x}"
+                    )
+        return out
+
     def pack_to_shm(self, t: "torch.Tensor"):
         """Like :meth:`pack_to_host` but the destination is a pooled
         /dev/shm segment (hipHostRegister-ed once) so a same-host peer can

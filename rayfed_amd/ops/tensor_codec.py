@@ -150,14 +150,25 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
             from rayfed_amd.ops import shm_pool
 
             if gpu_plane is not None and t.device.type == "cuda":
-                if (
+                wire_fp8 = (
                     gpu_plane.config.wire_dtype == "fp8e4m3"
                     and t.dtype == torch.bfloat16
-                ):
-                    man["wire"] = "fp8e4m3"
-                    man["nbytes"] = t.numel()
-                seg, crc, release = gpu_plane.pack_to_shm(t)
-                man["crc32"] = crc
+                )
+                if not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
+                    # Chunk pipeline: D2H publishes progress as it goes; the
+                    # frame leaves before the DMA finishes and the receiver's
+                    # H2D overlaps it.
+                    tc = t.detach()
+                    if not tc.is_contiguous():
+                        tc = tc.contiguous()
+                    seg, man_fields, release = gpu_plane.pack_to_shm_chunked(tc)
+                    man.update(man_fields)
+                else:
+                    if wire_fp8:
+                        man["wire"] = "fp8e4m3"
+                        man["nbytes"] = t.numel()
+                    seg, crc, release = gpu_plane.pack_to_shm(t)
+                    man["crc32"] = crc
             else:
                 pool = shm_pool.get_send_pool()
                 seg = pool.acquire(nbytes)
@@ -208,6 +219,55 @@ def release_parts(extras: Dict) -> None:
             pass
 
 
+def _decode_chunked(man: Dict, gpu_plane):
+    """Receiver of a chunk-pipelined shm push (see GpuDataPlane
+    .pack_to_shm_chunked).  GPU receivers overlap H2D with the sender's
+    in-flight D2H; CPU receivers wait for the final chunk then copy once."""
+    import struct
+    import time as _time
+
+    from rayfed_amd.ops import shm_pool
+
+    dtype = _STR_TO_DTYPE[man["dtype"]]
+    nbytes = man["nbytes"]
+    chunk = man["chunked"]
+    hdr = man["hdr"]
+    n_chunks = (nbytes + chunk - 1) // chunk
+    if gpu_plane is not None and man["device"] == "cuda" and gpu_plane.config.place_on_gpu:
+        return gpu_plane.unpack_from_shm_chunked(
+            man["shm"], man, dtype, man["shape"]
+        )
+    seg = shm_pool.attach(man["shm"])
+    deadline = _time.monotonic() + 600
+    while True:
+        ready = struct.unpack_from("<q", seg.array, 0)[0]
+        if ready >= n_chunks:
+            break
+        if ready < 0:
+            raise RuntimeError("peer aborted chunked shm push")
+        if _time.monotonic() > deadline:
+            raise TimeoutError("chunked shm push stalled")
+        _time.sleep(0.0005)
+    raw = seg.view(hdr, nbytes)
+    if man.get("crc_per_chunk"):
+        import zlib
+
+        for i in range(n_chunks):
+            lo, hi = i * chunk, min((i + 1) * chunk, nbytes)
+            want = struct.unpack_from("<I", seg.array, 8 + 4 * i)[0]
+            got = zlib.crc32(raw[lo:hi]) & 0xFFFFFFFF
+            if got != want:
+                raise ValueError(
+                    f"tensor CRC mismatch on chunk {i}: "
+                    f"expected {want:#x}, got {got:#x}"
+                )
+    return (
+        torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        .view(dtype)
+        .reshape(man["shape"])
+    )
+
+
 def decode(
     extras: Dict,
     payload: memoryview,
@@ -223,6 +283,9 @@ def decode(
     tensors: List[Any] = []
     for man in extras["tensors"]:
         nbytes = man["nbytes"]
+        if "chunked" in man:
+            tensors.append(_decode_chunked(man, gpu_plane))
+            continue
         if "shm" in man:
             from rayfed_amd.ops import shm_pool
 
