@@ -166,17 +166,190 @@ def _push_driver(party: str, addresses, steps: int, warmup: int, device: int,
     return elapsed
 
 
-_DRIVERS = {"tiny": _tiny_driver, "push": _push_driver}
+# ---------------------------------------------------------------- fedavg mode
+def llama3_8b_grad_shapes(layers: int = 32, vocab: int = 128256):
+    """Parameter shapes of Llama-3-8B (vocab 128256, d=4096, ffn 14336,
+    32 layers, GQA kv 1024) — synthetic gradients of this architecture.
+    ``vocab`` shrinks the embedding for small smoke runs only."""
+    shapes = [(vocab, 4096)]  # embed
+    for _ in range(layers):
+        shapes += [
+            (4096, 4096),   # q
+            (1024, 4096),   # k
+            (1024, 4096),   # v
+            (4096, 4096),   # o
+            (14336, 4096),  # gate
+            (14336, 4096),  # up
+            (4096, 14336),  # down
+            (4096,), (4096,),  # norms
+        ]
+    shapes += [(4096,), (vocab, 4096)]  # final norm, lm_head
+    return shapes
 
 
-def _run_single_process(mode, steps, warmup, push_bytes) -> float:
+def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
+                   vocab=128256, party_group=None, barrier_cb=None,
+                   result_q=None):
+    """Party-leader FedAvg round: (optional) intra-party RCCL all-reduce,
+    cross-party gradient exchange through the fed data plane, HIP weighted
+    combine (csrc fedavg_reduce_)."""
+    import torch
+
+    import rayfed_amd as fed
+    from rayfed_amd.parallel.fedavg import BucketedAllReducer, weighted_combine_
+
+    use_gpu = _has_cuda()
+    dev = f"cuda:{device}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(device)
+    fed.init(addresses=addresses, party=party, job_name=job_name,
+             logging_level="warning")
+
+    dtype = torch.bfloat16
+    torch.manual_seed(0 if party == "alice" else 1)
+    # empty+uniform_ is ~5x faster than randn for bf16 on CPU and identical
+    # for bandwidth purposes (content-independent copies/collectives).
+    grads = [
+        torch.empty(s, dtype=dtype, device=dev).uniform_(-1, 1)
+        for s in llama3_8b_grad_shapes(layers, vocab)
+    ]
+    nbytes = sum(g.numel() * g.element_size() for g in grads)
+    reducer = (
+        BucketedAllReducer(group=party_group) if party_group is not None else None
+    )
+    @fed.remote
+    class Exchanger:
+        """Receives the peer's averaged grads and combines with local."""
+
+        def combine(self, peer_flat, local_flat):
+            out = torch.empty_like(local_flat)
+            weighted_combine_(out, [local_flat, peer_flat], [0.5, 0.5])
+            if out.is_cuda:
+                torch.cuda.synchronize()
+            return float(out[:2].float().sum())
+
+    exchangers = {
+        "alice": Exchanger.party("alice").remote(),
+        "bob": Exchanger.party("bob").remote(),
+    }
+    # One flat buffer for the cross-party push (a single big tensor rides
+    # the chunk-pipelined shm lane).
+    flat = torch.empty(nbytes // 2, dtype=dtype, device=dev)
+
+    @fed.remote
+    def produce(_tick):
+        if reducer is not None:
+            reducer.allreduce_(grads)  # intra-party RCCL over xGMI
+        off = 0
+        for g in grads:
+            flat[off : off + g.numel()].copy_(g.view(-1))
+            off += g.numel()
+        return flat
+
+    def round_once(tick):
+        fa = produce.party("alice").remote(tick)
+        fb = produce.party("bob").remote(tick)
+        ca = exchangers["alice"].combine.remote(fb, fa)
+        cb = exchangers["bob"].combine.remote(fa, fb)
+        return fed.get([ca, cb])
+
+    for i in range(warmup):
+        round_once(("w", i))
+    if use_gpu:
+        torch.cuda.synchronize()
+    if barrier_cb is not None:
+        barrier_cb()
+    t0 = time.perf_counter()
+    for i in range(steps):
+        round_once(("s", i))
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if barrier_cb is not None:
+        barrier_cb()
+    fed.shutdown()
+    if result_q is not None:
+        result_q.put((elapsed, nbytes))
+    return elapsed, nbytes
+
+
+_DRIVERS = {"tiny": _tiny_driver, "push": _push_driver, "fedavg": _fedavg_driver}
+
+
+def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
+    """Leaders (ranks 0, W/2) run the fed exchange; every rank joins its
+    party's RCCL group for the intra-party gradient all-reduce."""
+    import torch
+    import torch.distributed as dist
+
+    from rayfed_amd.parallel.fedavg import BucketedAllReducer
+
+    dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+    half = world // 2
+    party = "alice" if rank < half else "bob"
+    use_gpu = _has_cuda()
+    backend = "nccl" if use_gpu else "gloo"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    alice_g = dist.new_group(list(range(half)), backend=backend)
+    bob_g = dist.new_group(list(range(half, world)), backend=backend)
+    group = alice_g if party == "alice" else bob_g
+    is_leader = rank in (0, half)
+    base = int(os.environ.get("RAYFED_BENCH_BASE_PORT", "23500"))
+    addresses = {
+        "alice": f"127.0.0.1:{base}",
+        "bob": f"127.0.0.1:{base + 1}",
+    }
+
+    def barrier():
+        dist.barrier()
+
+    if is_leader and half >= 1:
+        elapsed, nbytes = _fedavg_driver(
+            party, addresses, steps, warmup, local_rank, "bench_fedavg", layers,
+            vocab=vocab, party_group=group if half > 1 else None,
+            barrier_cb=barrier,
+        )
+    else:
+        # Member rank: mirror the leader's per-round intra-party all-reduce.
+        dev = f"cuda:{local_rank}" if use_gpu else "cpu"
+        torch.manual_seed(rank)
+        grads = [
+            torch.empty(s, dtype=torch.bfloat16, device=dev).uniform_(-1, 1)
+            for s in llama3_8b_grad_shapes(layers, vocab)
+        ]
+        nbytes = sum(g.numel() * g.element_size() for g in grads)
+        reducer = BucketedAllReducer(group=group)
+        for _ in range(warmup):
+            reducer.allreduce_(grads)
+        if use_gpu:
+            torch.cuda.synchronize()
+        barrier()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            reducer.allreduce_(grads)
+        if use_gpu:
+            torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        barrier()
+    t = __import__("torch").tensor([elapsed], dtype=__import__("torch").float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    dist.destroy_process_group()
+    return float(t.item()), nbytes
+
+
+def _run_single_process(mode, steps, warmup, extra=None):
     """N=1: alice in-process, bob forked, both on GPU 0."""
     from tests._util import make_addresses  # free-port helper
 
     addresses = make_addresses(["alice", "bob"])
     ctx = multiprocessing.get_context("fork")
-    kwargs = {}
-    args_extra = (push_bytes,) if mode == "push" else ()
+    if mode == "fedavg":
+        args_extra = tuple(extra)  # (layers, vocab)
+    elif mode == "push":
+        args_extra = (extra,)
+    else:
+        args_extra = ()
     bob = ctx.Process(
         target=_DRIVERS[mode],
         args=("bob", addresses, steps, warmup, 0, f"bench_{mode}") + args_extra,
@@ -227,35 +400,67 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=200)
-    p.add_argument("--mode", choices=["tiny", "push"], default="tiny")
+    p.add_argument("--mode", choices=["tiny", "push", "fedavg"], default="tiny")
     p.add_argument("--push-gib", type=float, default=4.0,
                    help="tensor size for --mode push (GiB)")
+    p.add_argument("--layers", type=int, default=32,
+                   help="transformer layers for --mode fedavg (32 = Llama-3-8B)")
+    p.add_argument("--vocab", type=int, default=128256,
+                   help="embedding vocab for --mode fedavg (smoke runs only)")
     args = p.parse_args()
 
-    if args.mode == "push" and args.steps > 50:
-        # 4 GiB per step: keep the default run under minutes.
-        args.steps = min(args.steps, 20)
-        args.warmup = min(args.warmup, 3)
+    if args.mode in ("push", "fedavg") and args.steps > 50:
+        # GiB-scale payload per step: keep the default run under minutes.
+        args.steps = min(args.steps, 10)
+        args.warmup = min(args.warmup, 2)
     push_bytes = int(args.push_gib * (1 << 30))
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
+    fedavg_nbytes = None
     if world > 1:
         assert world % 2 == 0, "world size must be even (2 parties)"
-        elapsed, lanes = _run_torchrun(
-            args.mode, args.steps, args.warmup, push_bytes, rank, world, local_rank
-        )
+        if args.mode == "fedavg":
+            elapsed, fedavg_nbytes = _run_fedavg_torchrun(
+                args.steps, args.warmup, args.layers, args.vocab, rank, world,
+                local_rank,
+            )
+            lanes = 1
+        else:
+            elapsed, lanes = _run_torchrun(
+                args.mode, args.steps, args.warmup, push_bytes, rank, world,
+                local_rank,
+            )
         if rank != 0:
             return
         n_gpus = world
     else:
-        elapsed = _run_single_process(args.mode, args.steps, args.warmup, push_bytes)
+        out = _run_single_process(
+            args.mode, args.steps, args.warmup,
+            (args.layers, args.vocab) if args.mode == "fedavg" else push_bytes,
+        )
+        if args.mode == "fedavg":
+            elapsed, fedavg_nbytes = out
+        else:
+            elapsed = out
         lanes, n_gpus = 1, args.gpus
 
     ms_per_step = elapsed * 1000.0 / args.steps
-    if args.mode == "tiny":
+    if args.mode == "fedavg":
+        # Gradient bytes exchanged cross-party per round (both directions).
+        value = 2.0 * fedavg_nbytes * args.steps / elapsed / 1e9
+        metric = "fedavg_cross_party_GBps"
+        unit = "GB/s"
+        config = {
+            "model": f"Llama-3-8B synthetic gradients ({args.layers} layers)",
+            "global_batch": args.steps,
+            "seq_len": fedavg_nbytes // 2,
+            "parallelism": f"2 parties x {max(1, world // 2)} GPUs, RCCL intra-party",
+            "grad_gib": round(fedavg_nbytes / (1 << 30), 2),
+        }
+    elif args.mode == "tiny":
         # 2 cross-party transfers per iteration per lane (value + broadcast).
         value = 2.0 * args.steps * lanes / elapsed
         metric = "cross_party_objects_per_sec"

@@ -196,3 +196,8 @@ def detach_all():
         _attach_cache.clear()
     for seg in segs:
         seg.close()
+
+
+# Close attachments before interpreter teardown so SharedMemory.__del__
+# never sees live numpy exports (noisy BufferError otherwise).
+atexit.register(detach_all)
