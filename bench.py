@@ -120,8 +120,11 @@ def _push_driver(party: str, addresses, steps: int, warmup: int, device: int,
     dev = f"cuda:{device}" if use_gpu else "cpu"
     if use_gpu:
         torch.cuda.set_device(device)
+    cfg = {}
+    if os.environ.get("RAYFED_BENCH_WIRE_FP8") == "1":
+        cfg["gpu_data_plane"] = {"wire_dtype": "fp8e4m3"}
     fed.init(addresses=addresses, party=party, job_name=job_name,
-             logging_level="warning")
+             config=cfg, logging_level="warning")
 
     numel = nbytes // 2
 
