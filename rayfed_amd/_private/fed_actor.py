@@ -23,6 +23,22 @@ def _invoke_materialized(fn, args, kwargs):
     return fn(*args, **kwargs)
 
 
+def _invoke_on_device(executor, device, fn, args, kwargs):
+    """Runs on a driver thread: materialize ObjectRef args locally, then
+    execute the call in GPU worker `device` and wait its result."""
+    args, kwargs = materialize((args, kwargs))
+    return executor.submit_on_device(device, fn, args, kwargs).result()
+
+
+# -- worker-side actor helpers (run inside the GPU worker process) ------------
+def _worker_create_actor(cls, args, kwargs):
+    return cls(*args, **kwargs)
+
+
+def _worker_call_method(instance, method_name, args, kwargs):
+    return getattr(instance, method_name)(*args, **kwargs)
+
+
 class FedActorHandle:
     def __init__(
         self,
@@ -62,15 +78,29 @@ class FedActorHandle:
 
     def _execute_impl(self, cls_args, cls_kwargs):
         """Create the real actor — only in the owning party
-        (reference fed_actor.py:78-91)."""
-        if self._node_party == self._party:
+        (reference fed_actor.py:78-91).  With ``options(device=k)`` the
+        actor lives in GPU worker process k, its state device-resident."""
+        if self._node_party != self._party:
+            return None
+        device = self._options.get("device") if self._options else None
+        if device is not None:
+            self._device = device
+            # Serialize device-actor calls via a 1-thread local actor that
+            # proxies into the worker (ordering parity with Ray actors).
             self._actor_handle = self._executor.create_actor(
-                _MaterializingActorFactory(self._body),
+                _DeviceActorShellFactory(
+                    self._executor, device, self._body
+                ),
                 args=(cls_args, cls_kwargs),
-                name=f"{self._body.__name__}-{self._fed_class_task_id}",
+                name=f"{self._body.__name__}-{self._fed_class_task_id}-dev{device}",
             )
             return self._actor_handle.ready_ref
-        return None
+        self._actor_handle = self._executor.create_actor(
+            _MaterializingActorFactory(self._body),
+            args=(cls_args, cls_kwargs),
+            name=f"{self._body.__name__}-{self._fed_class_task_id}",
+        )
+        return self._actor_handle.ready_ref
 
     def _execute_remote_method(self, method_name, options, args, kwargs):
         if self._actor_handle is None:
@@ -113,6 +143,41 @@ class _ActorShell:
     def call_method(self, method_name, args, kwargs):
         args, kwargs = materialize((args, kwargs))
         return getattr(self._instance, method_name)(*args, **kwargs)
+
+
+class _DeviceActorShellFactory:
+    """Creates the actor INSIDE GPU worker `device`; the driver-side shell
+    holds only a RemoteHandle and proxies method calls."""
+
+    __name__ = "_DeviceActorShellFactory"
+
+    def __init__(self, executor, device, body):
+        self._executor = executor
+        self._device = device
+        self._body = body
+
+    def __call__(self, cls_args, cls_kwargs):
+        args, kwargs = materialize((cls_args, cls_kwargs))
+        handle = self._executor.submit_on_device(
+            self._device, _worker_create_actor, (self._body, args, kwargs),
+            keep=True,
+        ).result()
+        return _DeviceActorShell(self._executor, self._device, handle)
+
+
+class _DeviceActorShell:
+    def __init__(self, executor, device, remote_handle):
+        self._executor = executor
+        self._device = device
+        self._handle = remote_handle
+
+    def call_method(self, method_name, args, kwargs):
+        args, kwargs = materialize((args, kwargs))
+        return self._executor.submit_on_device(
+            self._device,
+            _worker_call_method,
+            (self._handle, method_name, args, kwargs),
+        ).result()
 
 
 class FedActorMethod:

@@ -152,6 +152,20 @@ def init(
             use_global_proxy=use_global_proxy,
         )
 
+    # Per-GPU worker processes: config={"party_gpus": [0,1,2,3]} gives this
+    # party a slice of the node's MI355X GPUs — one worker process per GPU
+    # with a device-resident object table and an RCCL group over xGMI.
+    # Tasks/actors opt in via .options(device=k) (k indexes the slice).
+    party_gpus = config.get("party_gpus")
+    if party_gpus:
+        from rayfed_amd.runtime.worker import DeviceWorkerPool
+
+        pool = DeviceWorkerPool(
+            devices=list(party_gpus),
+            with_party_group=len(party_gpus) > 1,
+        )
+        ctx.get_executor().attach_worker_pool(pool)
+
     # Attach the GPU data plane when a HIP device is visible: tensors then
     # ride the pack/CRC/pinned-staging path instead of pickle.
     try:
@@ -254,6 +268,15 @@ class FedRemoteFunction:
 
     def _execute_impl(self, args, kwargs):
         ctx = get_global_context()
+        device = self._options.get("device")
+        if device is not None:
+            from rayfed_amd._private.fed_actor import _invoke_on_device
+
+            return ctx.get_executor().submit(
+                _invoke_on_device,
+                args=(ctx.get_executor(), device, self._func_body, args, kwargs),
+                num_returns=self._options.get("num_returns", 1),
+            )
         return ctx.get_executor().submit(
             _invoke_materialized,
             args=(self._func_body, args, kwargs),

@@ -50,6 +50,25 @@ class Executor:
         self._lock = threading.Lock()
         self._actors: List["ActorHandle"] = []
         self._shutdown = False
+        # Optional per-GPU worker processes (rayfed_amd.runtime.worker):
+        # tasks submitted with device=k route to worker k and their results
+        # can stay device-resident.
+        self._worker_pool = None
+
+    def attach_worker_pool(self, pool) -> None:
+        self._worker_pool = pool
+
+    @property
+    def worker_pool(self):
+        return self._worker_pool
+
+    def submit_on_device(self, device: int, fn, args=(), kwargs=None,
+                         keep: bool = False):
+        if self._worker_pool is None:
+            raise RuntimeError(
+                "no device worker pool attached; pass party_gpus to fed.init"
+            )
+        return self._worker_pool.submit(device, fn, tuple(args), kwargs, keep)
 
     # -- tasks ----------------------------------------------------------------
     def submit(
@@ -89,6 +108,9 @@ class Executor:
             self._actors.clear()
         for a in actors:
             a.kill(no_restart=True)
+        if self._worker_pool is not None:
+            self._worker_pool.shutdown()
+            self._worker_pool = None
         self._pool.shutdown(wait=wait)
 
 

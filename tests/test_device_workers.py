@@ -1,0 +1,106 @@
+"""fed API → per-device worker-process routing (.options(device=k)).
+
+CPU processes stand in for GPU workers here (devices=[None, ...]); on an
+MI355X node each worker pins one HIP device and joins the party's RCCL
+group (see test_parallel.py for the collective path).
+"""
+import os
+
+import rayfed_amd as fed
+from tests._util import run_parties
+
+
+def _driver_device_tasks(party, addresses):
+    fed.init(
+        addresses=addresses,
+        party=party,
+        config={"party_gpus": [None, None]},
+        logging_level="warning",
+    )
+
+    @fed.remote
+    def my_pid():
+        return os.getpid()
+
+    # Driver-local task vs device-worker tasks: distinct processes.
+    local = fed.get(my_pid.party(party).remote())
+    w0 = fed.get(my_pid.party(party).options(device=0).remote())
+    w1 = fed.get(my_pid.party(party).options(device=1).remote())
+    assert local == os.getpid()
+    assert w0 != local and w1 != local and w0 != w1
+
+    fed.shutdown()
+
+
+def test_device_tasks_run_in_worker_processes():
+    run_parties(_driver_device_tasks)
+
+
+def _driver_device_actor(party, addresses):
+    fed.init(
+        addresses=addresses,
+        party=party,
+        config={"party_gpus": [None]},
+        logging_level="warning",
+    )
+
+    @fed.remote
+    class Counter:
+        def __init__(self, start):
+            self.v = start
+            self.pid = os.getpid()
+
+        def add(self, n):
+            self.v += n
+            return (self.v, self.pid)
+
+    c = Counter.party("alice").options(device=0).remote(100)
+
+    @fed.remote
+    def check(pair, expect):
+        v, pid = pair
+        assert v == expect, (v, expect)
+        return pid
+
+    r1 = c.add.remote(1)
+    r2 = c.add.remote(2)  # state persists in the worker across calls
+    if party == "alice":
+        pid1 = fed.get(check.party("alice").remote(r1, 101))
+        pid2 = fed.get(check.party("alice").remote(r2, 103))
+        assert pid1 == pid2 != os.getpid()
+    else:
+        fed.get(check.party("alice").remote(r1, 101))
+        fed.get(check.party("alice").remote(r2, 103))
+    fed.shutdown()
+
+
+def test_device_actor_state_persists_in_worker():
+    run_parties(_driver_device_actor)
+
+
+def _driver_device_output_crosses_parties(party, addresses):
+    fed.init(
+        addresses=addresses,
+        party=party,
+        config={"party_gpus": [None]},
+        logging_level="warning",
+    )
+
+    @fed.remote
+    def produce():
+        import torch
+
+        return torch.arange(5, dtype=torch.float32)
+
+    @fed.remote
+    def total(t):
+        return float(t.sum())
+
+    o = produce.party("alice").options(device=0).remote()
+    r = total.party("bob").remote(o)
+    assert fed.get(r) == 10.0
+    fed.shutdown()
+
+
+def test_device_task_output_crosses_parties():
+    run_parties(_driver_device_output_crosses_parties)
