@@ -42,8 +42,8 @@ def _has_cuda():
 
 # ------------------------------------------------------------------ tiny mode
 def _tiny_driver(party: str, addresses, steps: int, warmup: int, device: int,
-                 job_name: str, barrier_cb=None, result_q=None):
-    """The many_tiny_tasks loop; identical code runs in both parties."""
+                 job_name: str, tls_config=None, barrier_cb=None, result_q=None):
+    """The many_tiny_tasks loop; identical code runs in every party."""
     import rayfed_amd as fed
 
     use_gpu = _has_cuda()
@@ -53,7 +53,8 @@ def _tiny_driver(party: str, addresses, steps: int, warmup: int, device: int,
         torch.cuda.set_device(device)
 
     fed.init(addresses=addresses, party=party, job_name=job_name,
-             logging_level="warning")
+             tls_config=tls_config, logging_level="warning")
+    parties = sorted(addresses)
 
     @fed.remote
     class MyActor:
@@ -71,21 +72,19 @@ def _tiny_driver(party: str, addresses, steps: int, warmup: int, device: int,
 
     @fed.remote
     class Aggregator:
-        def aggr(self, v1, v2):
-            return v1 + v2
+        def aggr(self, *vals):
+            return sum(vals)
 
-    actor_alice = MyActor.party("alice").remote(device, use_gpu)
-    actor_bob = MyActor.party("bob").remote(device, use_gpu)
-    aggregator = Aggregator.party("alice").remote()
+    actors = [MyActor.party(p).remote(device, use_gpu) for p in parties]
+    aggregator = Aggregator.party(parties[0]).remote()
 
     def one_iter():
-        va = actor_alice.run.remote()
-        vb = actor_bob.run.remote()
-        s = aggregator.aggr.remote(va, vb)
+        vals = [a.run.remote() for a in actors]
+        s = aggregator.aggr.remote(*vals)
         return fed.get(s)
 
     for _ in range(warmup):
-        assert one_iter() == 2
+        assert one_iter() == len(parties)
     if use_gpu:
         import torch
 
@@ -341,30 +340,46 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
     return float(t.item()), nbytes
 
 
-def _run_single_process(mode, steps, warmup, extra=None):
-    """N=1: alice in-process, bob forked, both on GPU 0."""
+def _run_single_process(mode, steps, warmup, extra=None, parties=2, tls=False):
+    """N=1: first party in-process, the rest forked, all on GPU 0."""
     from tests._util import make_addresses  # free-port helper
 
-    addresses = make_addresses(["alice", "bob"])
+    names = ["alice", "bob", "carol", "dave"][:parties]
+    addresses = make_addresses(names)
     ctx = multiprocessing.get_context("fork")
     if mode == "fedavg":
         args_extra = tuple(extra)  # (layers, vocab)
     elif mode == "push":
         args_extra = (extra,)
     else:
-        args_extra = ()
-    bob = ctx.Process(
-        target=_DRIVERS[mode],
-        args=("bob", addresses, steps, warmup, 0, f"bench_{mode}") + args_extra,
-    )
-    bob.start()
+        tls_config = None
+        if tls:
+            import sys as _sys
+
+            _sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+            from tool.generate_tls_certs import generate
+
+            tls_config = generate("/tmp/rayfed_amd/bench-certs")
+            tls_config["target_name_override"] = "localhost"
+        args_extra = (tls_config,)
+    procs = [
+        ctx.Process(
+            target=_DRIVERS[mode],
+            args=(name, addresses, steps, warmup, 0, f"bench_{mode}") + args_extra,
+            daemon=True,  # die with the driver — never orphan a party
+        )
+        for name in names[1:]
+    ]
+    for p in procs:
+        p.start()
     elapsed = _DRIVERS[mode](
-        "alice", addresses, steps, warmup, 0, f"bench_{mode}", *args_extra
+        names[0], addresses, steps, warmup, 0, f"bench_{mode}", *args_extra
     )
-    bob.join(timeout=600)
-    if bob.is_alive():
-        bob.terminate()
-        raise RuntimeError("bob party hung")
+    for p in procs:
+        p.join(timeout=600)
+        if p.is_alive():
+            p.terminate()
+            raise RuntimeError("peer party hung")
     return elapsed
 
 
@@ -410,6 +425,10 @@ def main():
                    help="transformer layers for --mode fedavg (32 = Llama-3-8B)")
     p.add_argument("--vocab", type=int, default=128256,
                    help="embedding vocab for --mode fedavg (smoke runs only)")
+    p.add_argument("--parties", type=int, default=2,
+                   help="party count for --mode tiny at N=1 (config 5 uses 3)")
+    p.add_argument("--tls", action="store_true",
+                   help="self-signed mutual TLS on the cross-silo link")
     args = p.parse_args()
 
     if args.mode in ("push", "fedavg") and args.steps > 50:
@@ -443,6 +462,8 @@ def main():
         out = _run_single_process(
             args.mode, args.steps, args.warmup,
             (args.layers, args.vocab) if args.mode == "fedavg" else push_bytes,
+            parties=args.parties if args.mode == "tiny" else 2,
+            tls=args.tls,
         )
         if args.mode == "fedavg":
             elapsed, fedavg_nbytes = out
@@ -451,6 +472,7 @@ def main():
         lanes, n_gpus = 1, args.gpus
 
     ms_per_step = elapsed * 1000.0 / args.steps
+    n_parties = args.parties if (args.mode == "tiny" and world <= 1) else 2
     if args.mode == "fedavg":
         # Gradient bytes exchanged cross-party per round (both directions).
         value = 2.0 * fedavg_nbytes * args.steps / elapsed / 1e9
@@ -464,12 +486,15 @@ def main():
             "grad_gib": round(fedavg_nbytes / (1 << 30), 2),
         }
     elif args.mode == "tiny":
-        # 2 cross-party transfers per iteration per lane (value + broadcast).
-        value = 2.0 * args.steps * lanes / elapsed
+        # Per iteration per lane: (P-1) value pushes into the aggregator's
+        # party + (P-1) result broadcasts on fed.get.
+        transfers = 2 * (n_parties - 1)
+        value = float(transfers) * args.steps * lanes / elapsed
         metric = "cross_party_objects_per_sec"
         unit = "objects/s"
         config = {
-            "model": "many_tiny_tasks (2-party aggregate loop)",
+            "model": f"many_tiny_tasks ({n_parties}-party aggregate loop"
+                     + (", TLS" if args.tls else "") + ")",
             "global_batch": args.steps * lanes,
             "seq_len": 1,
             "parallelism": f"fed2p-weak x{lanes} lanes",
