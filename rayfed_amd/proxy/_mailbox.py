@@ -66,6 +66,34 @@ class Mailbox:
         self._park(header, (self._KIND_OBJ, header, obj))
         return 200, "OK"
 
+    def try_take(self, upstream_seq_id, curr_seq_id):
+        """Opportunistic cross-thread take: return the parked item if its
+        delivery already completed, else None.  Safe off-loop: dict ops are
+        GIL-atomic, only the I/O loop creates/resolves slot futures, and a
+        done future is immutable."""
+        key = (str(upstream_seq_id), str(curr_seq_id))
+        fut = self._slots.get(key)
+        if fut is not None and fut.done() and not fut.cancelled():
+            self._slots.pop(key, None)
+            return fut.result()
+        return None
+
+    def consume_sync(self, item):
+        """Deserialize a taken item in the CALLING thread (fast path that
+        skips the I/O-loop round trip when data already arrived)."""
+        kind, header, payload = item
+        if kind == self._KIND_OBJ:
+            if isinstance(payload, BaseException):
+                raise payload
+            return payload
+        if kind == frames.KIND_ERROR:
+            raise serialization.loads(payload, self._allowed_list)
+        if kind == frames.KIND_TENSOR:
+            return tensor_codec.decode(
+                header, memoryview(payload), self.gpu_plane, self._allowed_list
+            )
+        return serialization.loads(payload, self._allowed_list)
+
     async def get_data(self, upstream_seq_id, curr_seq_id):
         key = (str(upstream_seq_id), str(curr_seq_id))
         fut = self._slots.get(key)

@@ -119,6 +119,37 @@ def get_service(name: str):
 # -----------------------------------------------------------------------------
 # Proxy services
 # -----------------------------------------------------------------------------
+class _EdgeStats:
+    """Per-destination transfer observability (the reference
+    exposes only op counters; we add bytes + latency)."""
+
+    __slots__ = ("ops", "bytes", "total_s", "max_s", "errors")
+
+    def __init__(self):
+        self.ops = 0
+        self.bytes = 0
+        self.total_s = 0.0
+        self.max_s = 0.0
+        self.errors = 0
+
+    def record(self, nbytes: int, secs: float, error: bool = False):
+        self.ops += 1
+        self.bytes += nbytes
+        self.total_s += secs
+        self.max_s = max(self.max_s, secs)
+        if error:
+            self.errors += 1
+
+    def as_dict(self):
+        return {
+            "ops": self.ops,
+            "bytes": self.bytes,
+            "avg_ms": (self.total_s / self.ops * 1e3) if self.ops else 0.0,
+            "max_ms": self.max_s * 1e3,
+            "errors": self.errors,
+        }
+
+
 class SenderProxyService:
     """Hosts a SenderProxy on the I/O loop; thread-safe send entry point."""
 
@@ -127,6 +158,7 @@ class SenderProxyService:
         self._io = io
         self._stats_lock = threading.Lock()
         self.send_op_count = 0
+        self._edges: Dict[str, _EdgeStats] = {}
 
     async def _send_coro(self, dest_party, data, upstream_seq_id, downstream_seq_id):
         if isinstance(data, ObjectRef):
@@ -135,9 +167,22 @@ class SenderProxyService:
             # FedRemoteError for the peer (reference barriers.py:147-174
             # gets the same effect from Ray arg resolution).
             data = await asyncio.wrap_future(data.future)
-        return await self._proxy.send(
-            dest_party, data, upstream_seq_id, downstream_seq_id
-        )
+        t0 = time.perf_counter()
+        try:
+            result = await self._proxy.send(
+                dest_party, data, upstream_seq_id, downstream_seq_id
+            )
+            err = False
+            return result
+        except BaseException:
+            err = True
+            raise
+        finally:
+            secs = time.perf_counter() - t0
+            with self._stats_lock:
+                edge = self._edges.setdefault(dest_party, _EdgeStats())
+                nbytes = getattr(self._proxy, "last_sent_bytes", 0)
+                edge.record(nbytes, secs, err)
 
     def send(self, dest_party, data, upstream_seq_id, downstream_seq_id) -> Future:
         with self._stats_lock:
@@ -146,8 +191,12 @@ class SenderProxyService:
             self._send_coro(dest_party, data, upstream_seq_id, downstream_seq_id)
         )
 
-    def _get_stats(self) -> Dict[str, int]:
-        return {"send_op_count": self.send_op_count}
+    def _get_stats(self) -> Dict[str, object]:
+        with self._stats_lock:
+            return {
+                "send_op_count": self.send_op_count,
+                "edges": {p: e.as_dict() for p, e in self._edges.items()},
+            }
 
     @property
     def proxy(self) -> SenderProxy:
@@ -175,6 +224,19 @@ class ReceiverProxyService:
     def get_data(self, src_party, upstream_seq_id, curr_seq_id) -> Future:
         with self._stats_lock:
             self.receive_op_count += 1
+        # Fast path: if the payload already landed, consume it right here in
+        # the caller's thread — no I/O-loop round trip (~0.1 ms saved on the
+        # tiny-task critical path).
+        mailbox = getattr(self._proxy, "_mailbox", None)
+        if mailbox is not None:
+            item = mailbox.try_take(upstream_seq_id, curr_seq_id)
+            if item is not None:
+                fut: Future = Future()
+                try:
+                    fut.set_result(mailbox.consume_sync(item))
+                except BaseException as e:  # noqa: BLE001
+                    fut.set_exception(e)
+                return fut
         return self._io.run_coro(
             self._proxy.get_data(src_party, upstream_seq_id, curr_seq_id)
         )

@@ -44,6 +44,7 @@ class GrpcSenderProxy(base_proxy.SenderProxy):
         self._channels: Dict[str, grpc.aio.Channel] = {}
         self._stubs: Dict[str, grpc.aio.UnaryUnaryMultiCallable] = {}
         self.gpu_plane = None  # attached by barriers when a GPU is present
+        self.last_sent_bytes = 0
         self._metadata = []
         if proxy_config is not None and getattr(proxy_config, "http_header", None):
             self._metadata = [
@@ -89,6 +90,7 @@ class GrpcSenderProxy(base_proxy.SenderProxy):
         stub = self._get_stub(dest_party)
         req = await self._encode_request(data, upstream_seq_id, downstream_seq_id)
         request = req.to_bytes()
+        self.last_sent_bytes = len(request)
         req.release()  # joined into one protobuf-free bytes body
         timeout = 60.0
         if self._proxy_config is not None and self._proxy_config.timeout_in_ms:

@@ -49,6 +49,10 @@ class TcpSenderReceiverProxy(SenderReceiverProxy):
     def received_op_count(self) -> int:
         return self._receiver.received_op_count
 
+    @property
+    def _mailbox(self):
+        return self._receiver._mailbox
+
     async def start(self):
         await self._receiver.start()
 

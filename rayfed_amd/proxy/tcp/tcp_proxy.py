@@ -167,6 +167,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
         self._conn_locks: Dict[str, asyncio.Lock] = {}
         self._same_host_cache: Dict[str, bool] = {}
         self.gpu_plane = None
+        self.last_sent_bytes = 0
         self._retry = _RetryPolicy(
             getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
         )
@@ -225,6 +226,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
             self._job_name, data, upstream_seq_id, downstream_seq_id,
             self.gpu_plane, shm=use_shm,
         )
+        self.last_sent_bytes = req.total_len
         try:
             return await self._send_framed(dest_party, req)
         finally:
