@@ -372,6 +372,73 @@ __global__ void fedavg_reduce_kernel(T* __restrict__ out, ReduceArgs args,
   }
 }
 
+// ---------------------------------------------------------------------------
+// MFMA variant of the weighted combine (bf16, k <= 32).
+//
+// Maps out[j] = Σ_k w_k·in_k[j] onto v_mfma_f32_16x16x32_bf16 as
+// C = A·B with A[i][kk] = in_kk[base+i] (16 output positions per MFMA) and
+// B[kk][j] = w_kk (constant across j).  C's column 0 lanes write the tile.
+//
+// This op is HBM-bound, so the VALU kernel above (16-byte lanes) is the
+// default and is faster — the guide's rule is that GEMM-shaped work belongs
+// on MFMA, and a weighted elementwise combine is not GEMM-shaped.  This
+// kernel exists as the measured MFMA alternative (numbers in
+// profiles/microbench): per MFMA only 16 of 256 output slots are useful and
+// the A-operand reads amplify 32/k×.
+// ---------------------------------------------------------------------------
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__global__ void fedavg_reduce_mfma_kernel(uint16_t* __restrict__ out,
+                                          ReduceArgs args,
+                                          unsigned long long n) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  const unsigned long long waves_total =
+      (unsigned long long)gridDim.x * blockDim.x / kWave;
+
+  // B fragment: lane l supplies B[kk][j] for kk = (l>>4)*8 + e, j = l&15.
+  bf16x8 b_frag;
+  {
+    const int kk_base = (lane >> 4) * 8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int kk = kk_base + e;
+      const float w = (kk < args.k) ? args.w[kk] : 0.0f;
+      b_frag[e] = (short)f32_to_bf16(w);
+    }
+  }
+
+  // Each wave iteration computes 16 consecutive outputs.
+  const int i = lane & 15;           // output position within the tile
+  const int kk_base = (lane >> 4) * 8;
+  for (unsigned long long base = (unsigned long long)wave * 16; base < n;
+       base += waves_total * 16) {
+    bf16x8 a_frag;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int kk = kk_base + e;
+      uint16_t v = 0;
+      if (kk < args.k && base + i < n) {
+        v = reinterpret_cast<const uint16_t*>(args.in[kk])[base + i];
+      }
+      a_frag[e] = (short)v;
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
+    // C layout: col = lane&15, row = (lane>>4)*4 + reg.  A's rows are the
+    // output positions, so lanes in column 0 write rows (= positions).
+    if ((lane & 15) == 0) {
+      const int row0 = (lane >> 4) * 4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const unsigned long long j = base + row0 + r;
+        if (j < n) out[j] = f32_to_bf16(acc[r]);
+      }
+    }
+  }
+}
+
 // dst += mask ? src : 0   (secure-aggregation style mask-add)
 template <typename T>
 __global__ void masked_add_kernel(T* __restrict__ dst,
@@ -565,6 +632,33 @@ void fedavg_reduce_(torch::Tensor out, std::vector<torch::Tensor> inputs,
   }
 }
 
+void fedavg_reduce_mfma_(torch::Tensor out, std::vector<torch::Tensor> inputs,
+                         std::vector<double> weights) {
+  TORCH_CHECK(!inputs.empty() && inputs.size() <= 32,
+              "fedavg_reduce_mfma_: 1..32 inputs");
+  TORCH_CHECK(weights.size() == inputs.size(), "weights/inputs mismatch");
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() &&
+                  out.dtype() == torch::kBFloat16,
+              "out must be contiguous CUDA bf16");
+  const unsigned long long n = out.numel();
+  ReduceArgs args;
+  args.k = (int)inputs.size();
+  for (int j = 0; j < args.k && j < kMaxInputs; ++j) {
+    TORCH_CHECK(inputs[j].is_cuda() && inputs[j].is_contiguous() &&
+                    inputs[j].numel() == (long long)n &&
+                    inputs[j].dtype() == out.dtype(),
+                "input ", j, " mismatch");
+    args.in[j] = inputs[j].data_ptr();
+    args.w[j] = (float)weights[j];
+  }
+  TORCH_CHECK(args.k <= kMaxInputs,
+              "fedavg_reduce_mfma_: pointer table limited to ", kMaxInputs);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fedavg_reduce_mfma_kernel, dim3(4096), dim3(kBlock), 0,
+                     stream, reinterpret_cast<uint16_t*>(out.data_ptr()), args,
+                     n);
+}
+
 void masked_add_(torch::Tensor dst, torch::Tensor src, torch::Tensor mask) {
   TORCH_CHECK(dst.is_cuda() && dst.is_contiguous() && src.is_contiguous() &&
                   mask.is_contiguous(),
@@ -622,6 +716,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("unpack_fp8_async", &unpack_fp8_async, "fp8 e4m3 bytes -> bf16");
   m.def("fedavg_reduce_", &fedavg_reduce_,
         "out = sum_k w_k * in_k (bf16/f16/f32, fp32 accumulation)");
+  m.def("fedavg_reduce_mfma_", &fedavg_reduce_mfma_,
+        "MFMA (v_mfma_f32_16x16x32_bf16) variant of the weighted combine");
   m.def("masked_add_", &masked_add_, "dst += mask ? src : 0");
   m.def("crc32_combine", &crc32_combine_py, "zlib-style CRC combine");
   m.def("host_register", &host_register, "hipHostRegister an existing mapping");

@@ -79,6 +79,19 @@ def main():
         moved = (k + 1) * n * 2
         report(f"fedavg_reduce k={k} bf16", moved, timeit(run), {"elems": n})
 
+    # MFMA combine variant (documented comparison; VALU path is default)
+    for k in [4, 16]:
+        n = 1 << 28
+        ins = [torch.randn(n, device="cuda").to(torch.bfloat16) for _ in range(k)]
+        outt = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+        w = [1.0 / k] * k
+
+        def run_mfma():
+            ext.fedavg_reduce_mfma_(outt, ins, w)
+
+        report(f"fedavg_reduce_MFMA k={k} bf16", (k + 1) * n * 2,
+               timeit(run_mfma), {"elems": n})
+
     # DMA: pinned D2H / H2D
     n = 1 << 30
     dev = torch.empty(n, dtype=torch.uint8, device="cuda")

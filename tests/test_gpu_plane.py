@@ -278,3 +278,23 @@ def test_chunked_shm_crc_tamper(plane):
         tensor_codec.decode(extras, memoryview(b""), plane, None)
     tensor_codec.release_parts(extras)
     shm_pool.detach_all()
+
+
+@needs_gpu
+def test_fedavg_reduce_mfma_matches_reference(ext):
+    """MFMA weighted-combine variant vs the fp32 torch reference — same
+    numerics contract as the VALU kernel (bf16 in, bf16 out).  The MFMA
+    internal accumulation is fp32, but the A/B operands are the bf16 inputs
+    and bf16(w), so allow one extra bf16 rounding on w."""
+    torch.manual_seed(3)
+    n = 1_000_003
+    for k in (2, 4, 7):
+        ins = [torch.randn(n, device="cuda").to(torch.bfloat16) for _ in range(k)]
+        w = [(i + 1) / sum(range(1, k + 1)) for i in range(k)]
+        out = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+        ext.fedavg_reduce_mfma_(out, ins, w)
+        torch.cuda.synchronize()
+        wq = [float(torch.tensor(x).to(torch.bfloat16).float()) for x in w]
+        ref = sum(wi * x.float() for wi, x in zip(wq, ins)).to(torch.bfloat16)
+        diff = (out.float() - ref.float()).abs().max().item()
+        assert diff <= 0.06, f"k={k}: max diff {diff}"
