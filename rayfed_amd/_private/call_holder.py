@@ -16,7 +16,7 @@ both parties name the same call identically without negotiation.
 from __future__ import annotations
 
 import logging
-from typing import Any, Callable, Optional
+from typing import Callable, Optional
 
 from rayfed_amd import tree_util
 from rayfed_amd._private.global_context import get_global_context

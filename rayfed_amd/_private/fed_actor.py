@@ -9,7 +9,7 @@ actor).
 from __future__ import annotations
 
 import logging
-from typing import Any, Dict, Optional
+from typing import Dict, Optional
 
 from rayfed_amd._private.call_holder import FedCallHolder
 from rayfed_amd.runtime.executor import ActorHandle, Executor

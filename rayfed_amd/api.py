@@ -14,7 +14,7 @@ import logging
 import signal
 import sys
 import threading
-from typing import Any, Callable, Dict, List, Optional, Union
+from typing import Any, Callable, Dict, List, Union
 
 import cloudpickle
 

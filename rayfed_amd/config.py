@@ -11,7 +11,7 @@ HIP pack → pinned-staging → gRPC pipeline and CRC verification toggles.
 from __future__ import annotations
 
 import json
-from dataclasses import dataclass, field, fields
+from dataclasses import dataclass, fields
 from typing import Dict, List, Optional
 
 import cloudpickle

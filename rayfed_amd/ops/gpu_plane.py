@@ -24,7 +24,7 @@ from __future__ import annotations
 
 import logging
 import threading
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from rayfed_amd.config import GpuDataPlaneConfig
 

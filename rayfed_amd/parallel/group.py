@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import datetime
 import logging
-import os
 from typing import Optional
 
 import torch

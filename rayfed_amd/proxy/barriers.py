@@ -26,7 +26,7 @@ from typing import Any, Dict, Optional
 from rayfed_amd._private import constants
 from rayfed_amd._private.global_context import get_global_context
 from rayfed_amd.config import CrossSiloMessageConfig
-from rayfed_amd.proxy.base_proxy import ReceiverProxy, SenderProxy, SenderReceiverProxy
+from rayfed_amd.proxy.base_proxy import ReceiverProxy, SenderProxy
 from rayfed_amd.runtime.object_ref import ObjectRef
 
 logger = logging.getLogger(__name__)

@@ -28,7 +28,7 @@ Responses are msgpack ``{code: int, result: str}`` with HTTP-style codes
 from __future__ import annotations
 
 import struct
-from typing import Any, Dict, Optional, Tuple, Union
+from typing import Any, Dict, Tuple, Union
 
 import msgpack
 

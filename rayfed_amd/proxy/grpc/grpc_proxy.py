@@ -16,16 +16,12 @@ _run_grpc_server :345-381) with these deliberate differences:
 """
 from __future__ import annotations
 
-import asyncio
 import logging
-from typing import Any, Dict, Optional, Tuple
+from typing import Dict, Optional
 
 import grpc
 
 from rayfed_amd import config as fed_config
-from rayfed_amd._private import serialization
-from rayfed_amd.exceptions import FedRemoteError
-from rayfed_amd.ops import tensor_codec
 from rayfed_amd.proxy import base_proxy
 from rayfed_amd.proxy._mailbox import Mailbox
 from rayfed_amd.proxy.grpc import frames, grpc_options

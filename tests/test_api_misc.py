@@ -6,7 +6,7 @@ import multiprocessing
 import pytest
 
 import rayfed_amd as fed
-from tests._util import free_ports, make_addresses, run_parties
+from tests._util import make_addresses, run_parties
 
 _mp = multiprocessing.get_context("fork")
 

@@ -1,7 +1,6 @@
 """Config dataclasses + internal KV (parity: reference test_api.py config
 readback, test_internal_kv.py, test_retry_policy.py plumb-through)."""
 import cloudpickle
-import pytest
 
 from rayfed_amd import config as fed_config
 from rayfed_amd._private import constants, kv as kv_mod

@@ -1,6 +1,4 @@
 """Same-host shared-memory lane (CPU paths; GPU variants in test_gpu_plane)."""
-import multiprocessing
-
 import pytest
 
 torch = pytest.importorskip("torch")

@@ -3,7 +3,6 @@
 GPU variants (HIP pack kernel / pinned staging) live in test_gpu_plane.py
 and are marked ``gpu``.
 """
-import numpy as np
 import pytest
 
 torch = pytest.importorskip("torch")

@@ -8,7 +8,6 @@ from rayfed_amd._private.global_context import (
     clear_global_context,
     init_global_context,
 )
-from rayfed_amd.config import GrpcCrossSiloMessageConfig
 from tests._util import make_addresses
 
 
