@@ -1,0 +1,557 @@
+// rayfed_amd C++ transport core — the cross-silo hot path without asyncio.
+//
+// The reference's data plane is Python gRPC inside Ray actor processes
+// (/root/reference/fed/proxy/grpc/grpc_proxy.py); this engine's default
+// Python transport is framed asyncio TCP (~0.1 ms/op).  This module is the
+// native tier (SURVEY.md §2.3 "C++ data plane for the proxy hot path"):
+// persistent sockets, framing, routing, the receive mailbox and ack
+// round-trips all run in C++ threads with the GIL released — Python only
+// (de)serializes payloads at the edges.
+//
+// Wire format (little-endian), the "xfer" framing:
+//   u64 total_len ‖ u64 req_id ‖ u8 flags ‖ u8 job_len ‖ u8 up_len ‖
+//   u8 down_len ‖ job ‖ up ‖ down ‖ body (opaque to C++)
+//   ack: u32 len ‖ u64 req_id ‖ u16 code ‖ result (utf-8)
+// flags bit 0 (DEFER_ACK): consume via the Python callback BEFORE acking
+// (the shm lane's ack licenses segment recycling).
+//
+// Scope: plaintext only — TLS jobs use the Python asyncio transport
+// (selected automatically in rayfed_amd.proxy.barriers).  Thread-per-
+// connection: a federation has a handful of parties, not thousands.
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <arpa/inet.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <sys/uio.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <condition_variable>
+#include <cstring>
+#include <deque>
+#include <map>
+#include <mutex>
+#include <optional>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+constexpr uint8_t kFlagDeferAck = 1;
+
+// ---------------------------------------------------------------- utilities
+static void write_all(int fd, const char* data, size_t n) {
+  while (n) {
+    ssize_t w = ::send(fd, data, n, MSG_NOSIGNAL);
+    if (w < 0) {
+      if (errno == EINTR) continue;
+      throw std::runtime_error(std::string("send: ") + strerror(errno));
+    }
+    data += w;
+    n -= (size_t)w;
+  }
+}
+
+static void writev_all(int fd, std::vector<iovec> iov) {
+  size_t idx = 0;
+  while (idx < iov.size()) {
+    ssize_t w = ::writev(fd, iov.data() + idx, (int)(iov.size() - idx));
+    if (w < 0) {
+      if (errno == EINTR) continue;
+      throw std::runtime_error(std::string("writev: ") + strerror(errno));
+    }
+    size_t ww = (size_t)w;
+    while (idx < iov.size() && ww >= iov[idx].iov_len) {
+      ww -= iov[idx].iov_len;
+      ++idx;
+    }
+    if (idx < iov.size() && ww) {
+      iov[idx].iov_base = (char*)iov[idx].iov_base + ww;
+      iov[idx].iov_len -= ww;
+    }
+  }
+}
+
+static bool read_all(int fd, char* data, size_t n) {
+  while (n) {
+    ssize_t r = ::recv(fd, data, n, 0);
+    if (r < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    if (r == 0) return false;  // peer closed
+    data += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+static void set_sock_opts(int fd) {
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  int buf = 8 << 20;
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof(buf));
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
+}
+
+struct Ack {
+  uint16_t code;
+  std::string result;
+};
+
+// ------------------------------------------------------------------- server
+class XferServer {
+ public:
+  XferServer(int port, std::string job_name)
+      : job_(std::move(job_name)), port_(port) {}
+
+  ~XferServer() { stop(); }
+
+  // consume_cb(body_bytes) -> int code; called (with the GIL) for
+  // DEFER_ACK frames only.  Delivery happens before the ack is written.
+  void start(py::object consume_cb) {
+    consume_cb_ = std::move(consume_cb);
+    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+    int zero = 0, one = 1;
+    // Parity with grpc.so_reuseport=0: exclusive bind, but allow
+    // REUSEADDR so TIME_WAIT ports rebind across test runs.
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEPORT, &zero, sizeof(zero));
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = htonl(INADDR_ANY);
+    addr.sin_port = htons((uint16_t)port_);
+    if (bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0) {
+      ::close(listen_fd_);
+      listen_fd_ = -1;
+      throw std::runtime_error("bind failed: port in use");
+    }
+    if (listen(listen_fd_, 64) != 0) {
+      ::close(listen_fd_);
+      listen_fd_ = -1;
+      throw std::runtime_error("listen failed");
+    }
+    running_ = true;
+    accept_thread_ = std::thread([this] { accept_loop(); });
+  }
+
+  void stop() {
+    bool expected = true;
+    if (!running_.compare_exchange_strong(expected, false)) return;
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    {
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+    }
+    if (accept_thread_.joinable()) accept_thread_.join();
+    {
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      for (auto& t : conn_threads_)
+        if (t.joinable()) t.join();
+      conn_threads_.clear();
+    }
+    // Unblock any waiting get_data.
+    std::lock_guard<std::mutex> lk(mail_mu_);
+    stopped_ = true;
+    mail_cv_.notify_all();
+  }
+
+  // Blocking fetch (GIL released by the binding); returns body bytes.
+  py::bytes get_data(const std::string& up, const std::string& down,
+                     double timeout_s) {
+    std::string key = up + '\x00' + down;
+    std::string body;
+    {
+      py::gil_scoped_release release;
+      std::unique_lock<std::mutex> lk(mail_mu_);
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::duration<double>(timeout_s);
+      while (true) {
+        auto it = mail_.find(key);
+        if (it != mail_.end()) {
+          body = std::move(it->second);
+          mail_.erase(it);
+          break;
+        }
+        if (stopped_) throw std::runtime_error("server stopped");
+        if (mail_cv_.wait_until(lk, deadline) == std::cv_status::timeout)
+          throw std::runtime_error("get_data timeout");
+      }
+    }
+    return py::bytes(body);
+  }
+
+  // Insert a marker/body locally (used after a deferred consume so a
+  // blocked get_data wakes and finds the Python-side decoded object).
+  void post(const std::string& up, const std::string& down,
+            const std::string& body) {
+    std::lock_guard<std::mutex> lk(mail_mu_);
+    mail_[up + '\x00' + down] = body;
+    mail_cv_.notify_all();
+  }
+
+  // Non-blocking probe for the fast path.
+  std::optional<py::bytes> try_take(const std::string& up,
+                                    const std::string& down) {
+    std::string key = up + '\x00' + down;
+    std::string body;
+    {
+      py::gil_scoped_release release;
+      std::lock_guard<std::mutex> lk(mail_mu_);
+      auto it = mail_.find(key);
+      if (it == mail_.end()) return std::nullopt;
+      body = std::move(it->second);
+      mail_.erase(it);
+    }
+    return py::bytes(body);
+  }
+
+  uint64_t received_op_count() const { return recv_count_.load(); }
+
+ private:
+  void accept_loop() {
+    while (running_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (running_ && errno == EINTR) continue;
+        break;
+      }
+      set_sock_opts(fd);
+      std::lock_guard<std::mutex> lk(conn_mu_);
+      conn_fds_.push_back(fd);
+      conn_threads_.emplace_back([this, fd] { conn_loop(fd); });
+    }
+  }
+
+  void conn_loop(int fd) {
+    std::vector<char> buf;
+    while (running_) {
+      char head[16];
+      if (!read_all(fd, head, 16)) break;
+      uint64_t total, req_id;
+      memcpy(&total, head, 8);
+      memcpy(&req_id, head + 8, 8);
+      if (total < 12 || total > (64ull << 30)) break;  // sane bounds
+      buf.resize(total - 8);  // everything after req_id
+      if (!read_all(fd, buf.data(), buf.size())) break;
+      uint8_t flags = (uint8_t)buf[0];
+      uint8_t job_len = (uint8_t)buf[1];
+      uint8_t up_len = (uint8_t)buf[2];
+      uint8_t down_len = (uint8_t)buf[3];
+      size_t off = 4;
+      if (off + job_len + up_len + down_len > buf.size()) break;
+      std::string job(buf.data() + off, job_len);
+      off += job_len;
+      std::string up(buf.data() + off, up_len);
+      off += up_len;
+      std::string down(buf.data() + off, down_len);
+      off += down_len;
+
+      uint16_t code = 200;
+      std::string result = "OK";
+      if (job != job_) {
+        code = 417;
+        result = "JobName mis-match: expected " + job_ + ", got " + job;
+      } else if (flags & kFlagDeferAck) {
+        // shm-lane frame: Python consumes (H2D + CRC) before we ack.
+        py::gil_scoped_acquire gil;
+        try {
+          py::bytes body(buf.data() + off, buf.size() - off);
+          code = (uint16_t)py::cast<int>(consume_cb_(up, down, body));
+        } catch (const std::exception& e) {
+          code = 500;
+          result = std::string("consume failed: ") + e.what();
+        }
+      } else {
+        recv_count_.fetch_add(1);
+        std::lock_guard<std::mutex> lk(mail_mu_);
+        mail_[up + '\x00' + down] =
+            std::string(buf.data() + off, buf.size() - off);
+        mail_cv_.notify_all();
+      }
+
+      char ack[14];
+      uint32_t ack_len = 10 + (uint32_t)result.size();
+      memcpy(ack, &ack_len, 4);
+      memcpy(ack + 4, &req_id, 8);
+      memcpy(ack + 12, &code, 2);
+      try {
+        write_all(fd, ack, 14);
+        if (!result.empty()) write_all(fd, result.data(), result.size());
+      } catch (...) {
+        break;
+      }
+    }
+    ::close(fd);
+  }
+
+  std::string job_;
+  int port_;
+  int listen_fd_ = -1;
+  std::atomic<bool> running_{false};
+  bool stopped_ = false;
+  std::thread accept_thread_;
+  std::mutex conn_mu_;
+  std::vector<int> conn_fds_;
+  std::vector<std::thread> conn_threads_;
+  std::mutex mail_mu_;
+  std::condition_variable mail_cv_;
+  std::map<std::string, std::string> mail_;
+  std::atomic<uint64_t> recv_count_{0};
+  py::object consume_cb_;
+};
+
+// ------------------------------------------------------------------- client
+class XferClient {
+ public:
+  XferClient(std::string job_name) : job_(std::move(job_name)) {}
+  ~XferClient() { close_all(); }
+
+  // Blocking send with ack round trip; GIL released around I/O.
+  // parts: list of buffer-likes written scatter-gather (no join copy).
+  int send(const std::string& host, int port, const std::string& up,
+           const std::string& down, std::vector<py::buffer> parts,
+           bool defer_ack, double timeout_s, std::string* result_out) {
+    // Collect buffer pointers under the GIL.
+    std::vector<std::pair<const char*, size_t>> views;
+    views.reserve(parts.size());
+    size_t body_len = 0;
+    std::vector<py::buffer_info> infos;
+    infos.reserve(parts.size());
+    for (auto& b : parts) {
+      infos.emplace_back(b.request());
+      auto& info = infos.back();
+      views.emplace_back((const char*)info.ptr,
+                         (size_t)(info.size * info.itemsize));
+      body_len += views.back().second;
+    }
+
+    uint8_t flags = defer_ack ? kFlagDeferAck : 0;
+    std::string preamble;
+    preamble.push_back((char)flags);
+    preamble.push_back((char)job_.size());
+    preamble.push_back((char)up.size());
+    preamble.push_back((char)down.size());
+    preamble += job_;
+    preamble += up;
+    preamble += down;
+
+    uint64_t total = 8 /*req id*/ + preamble.size() + body_len;
+
+    int code;
+    std::string result;
+    {
+      py::gil_scoped_release release;
+      std::shared_ptr<Conn> conn_sp = get_conn(host, port);
+      Conn& conn = *conn_sp;  // shared ownership: safe vs concurrent replace
+      uint64_t req_id;
+      std::shared_ptr<Pending> pending = std::make_shared<Pending>();
+      {
+        std::lock_guard<std::mutex> lk(conn.write_mu);
+        req_id = conn.next_id++;
+        {
+          std::lock_guard<std::mutex> lk2(conn.pend_mu);
+          conn.pending[req_id] = pending;
+        }
+        char head[16];
+        memcpy(head, &total, 8);
+        memcpy(head + 8, &req_id, 8);
+        std::vector<iovec> iov;
+        iov.push_back({head, 16});
+        iov.push_back({(void*)preamble.data(), preamble.size()});
+        for (auto& v : views) iov.push_back({(void*)v.first, v.second});
+        try {
+          writev_all(conn.fd, std::move(iov));
+        } catch (...) {
+          conn.alive = false;
+          throw;
+        }
+      }
+      std::unique_lock<std::mutex> lk(pending->mu);
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::duration<double>(timeout_s);
+      while (!pending->done) {
+        if (pending->cv.wait_until(lk, deadline) == std::cv_status::timeout) {
+          throw std::runtime_error("ack timeout");
+        }
+      }
+      if (pending->broken)
+        throw std::runtime_error("connection broken awaiting ack");
+      code = pending->code;
+      result = pending->result;
+    }
+    if (result_out) *result_out = result;
+    return code;
+  }
+
+  void close_all() {
+    std::unordered_map<std::string, std::shared_ptr<Conn>> conns;
+    {
+      std::lock_guard<std::mutex> lk(conns_mu_);
+      conns.swap(conns_);
+    }
+    for (auto& [key, conn] : conns) {
+      ::shutdown(conn->fd, SHUT_RDWR);
+      if (conn->reader.joinable()) conn->reader.join();
+      ::close(conn->fd);
+    }
+  }
+
+ private:
+  struct Pending {
+    std::mutex mu;
+    std::condition_variable cv;
+    bool done = false;
+    bool broken = false;
+    uint16_t code = 0;
+    std::string result;
+  };
+
+  struct Conn {
+    int fd = -1;
+    uint64_t next_id = 1;
+    std::mutex write_mu;
+    std::mutex pend_mu;
+    std::unordered_map<uint64_t, std::shared_ptr<Pending>> pending;
+    std::thread reader;
+    std::atomic<bool> alive{true};
+  };
+
+  std::shared_ptr<Conn> get_conn(const std::string& host, int port) {
+    std::string key = host + ":" + std::to_string(port);
+    // Serialize (re)connection per client: the lock is held through
+    // connect() so concurrent first-sends share one socket, and dead conns
+    // are REPLACED, never destroyed, while a sender still references them
+    // (shared_ptr ownership).
+    std::lock_guard<std::mutex> lk(conns_mu_);
+    auto it = conns_.find(key);
+    if (it != conns_.end() && it->second->alive) return it->second;
+    int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (fd < 0) throw std::runtime_error("socket() failed");
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_port = htons((uint16_t)port);
+    if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1) {
+      hostent* he = gethostbyname(host.c_str());
+      if (!he) {
+        ::close(fd);
+        throw std::runtime_error("resolve failed: " + host);
+      }
+      memcpy(&addr.sin_addr, he->h_addr, sizeof(addr.sin_addr));
+    }
+    if (connect(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
+      ::close(fd);
+      throw std::runtime_error(std::string("connect failed: ") +
+                               strerror(errno));
+    }
+    set_sock_opts(fd);
+    auto conn = std::make_shared<Conn>();
+    conn->fd = fd;
+    std::shared_ptr<Conn> sp = conn;
+    conn->reader = std::thread([this, sp] { reader_loop(sp.get()); });
+    auto& slot = conns_[key];
+    if (slot) {  // replace a dead conn; readers/senders keep their refs
+      ::shutdown(slot->fd, SHUT_RDWR);
+      if (slot->reader.joinable()) slot->reader.detach();
+      dead_.push_back(slot);
+    }
+    slot = conn;
+    return conn;
+  }
+
+  void reader_loop(Conn* conn) {
+    while (true) {
+      char head[14];
+      if (!read_all(conn->fd, head, 14)) break;
+      uint32_t len;
+      uint64_t req_id;
+      uint16_t code;
+      memcpy(&len, head, 4);
+      memcpy(&req_id, head + 4, 8);
+      memcpy(&code, head + 12, 2);
+      std::string result;
+      if (len > 10) {
+        result.resize(len - 10);
+        if (!read_all(conn->fd, result.data(), result.size())) break;
+      }
+      std::shared_ptr<Pending> p;
+      {
+        std::lock_guard<std::mutex> lk(conn->pend_mu);
+        auto it = conn->pending.find(req_id);
+        if (it != conn->pending.end()) {
+          p = it->second;
+          conn->pending.erase(it);
+        }
+      }
+      if (p) {
+        std::lock_guard<std::mutex> lk(p->mu);
+        p->done = true;
+        p->code = code;
+        p->result = std::move(result);
+        p->cv.notify_all();
+      }
+    }
+    conn->alive = false;
+    std::lock_guard<std::mutex> lk(conn->pend_mu);
+    for (auto& [id, p] : conn->pending) {
+      std::lock_guard<std::mutex> lk2(p->mu);
+      p->done = true;
+      p->broken = true;
+      p->cv.notify_all();
+    }
+    conn->pending.clear();
+  }
+
+  std::string job_;
+  std::mutex conns_mu_;
+  std::unordered_map<std::string, std::shared_ptr<Conn>> conns_;
+  std::vector<std::shared_ptr<Conn>> dead_;  // kept until close_all
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_xfer, m) {
+  m.doc() = "rayfed_amd C++ transport core (plaintext cross-silo hot path)";
+  py::class_<XferServer>(m, "XferServer")
+      .def(py::init<int, std::string>(), py::arg("port"), py::arg("job_name"))
+      .def("start", &XferServer::start, py::arg("consume_cb"))
+      .def("stop", &XferServer::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def("get_data", &XferServer::get_data, py::arg("up"), py::arg("down"),
+           py::arg("timeout_s") = 600.0)
+      .def("post", &XferServer::post, py::call_guard<py::gil_scoped_release>())
+      .def("try_take", &XferServer::try_take)
+      .def_property_readonly("received_op_count",
+                             &XferServer::received_op_count);
+  py::class_<XferClient>(m, "XferClient")
+      .def(py::init<std::string>(), py::arg("job_name"))
+      .def(
+          "send",
+          [](XferClient& c, const std::string& host, int port,
+             const std::string& up, const std::string& down,
+             std::vector<py::buffer> parts, bool defer_ack,
+             double timeout_s) {
+            std::string result;
+            int code =
+                c.send(host, port, up, down, std::move(parts), defer_ack,
+                       timeout_s, &result);
+            return py::make_tuple(code, result);
+          },
+          py::arg("host"), py::arg("port"), py::arg("up"), py::arg("down"),
+          py::arg("parts"), py::arg("defer_ack") = false,
+          py::arg("timeout_s") = 60.0)
+      .def("close_all", &XferClient::close_all,
+           py::call_guard<py::gil_scoped_release>());
+}

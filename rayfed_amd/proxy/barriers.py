@@ -270,6 +270,26 @@ def _get_io_loop() -> IoLoop:
     return _io_loop
 
 
+def _use_cpp_transport(tls_config) -> bool:
+    """Default to the C++ transport core for plaintext jobs when built.
+    TLS rides the Python asyncio transport (ssl-integrated); RAYFED_TRANSPORT
+    ∈ {cpp, asyncio} forces a choice."""
+    import os
+
+    mode = os.environ.get("RAYFED_TRANSPORT", "auto")
+    if mode == "asyncio":
+        return False
+    if tls_config:
+        return False
+    from rayfed_amd.proxy.xfer import xfer_available
+
+    if mode == "cpp":
+        if not xfer_available():
+            raise RuntimeError("RAYFED_TRANSPORT=cpp but rayfed_amd._xfer missing")
+        return True
+    return xfer_available()
+
+
 def start_receiver_proxy(
     addresses: Dict,
     party: str,
@@ -281,6 +301,15 @@ def start_receiver_proxy(
     use_global_proxy: bool = True,
 ):
     global _receiver_service
+    if proxy_cls is None and _use_cpp_transport(tls_config):
+        from rayfed_amd.proxy.xfer import XferReceiverService
+
+        service = XferReceiverService(
+            addresses[party], party, job_name, proxy_config
+        )
+        _receiver_service = service
+        _service_registry[receiver_proxy_name(job_name, use_global_proxy)] = service
+        return service
     if proxy_cls is None:
         from rayfed_amd.proxy.tcp.tcp_proxy import TcpReceiverProxy
 
@@ -307,6 +336,13 @@ def start_sender_proxy(
     use_global_proxy: bool = True,
 ):
     global _sender_service
+    if proxy_cls is None and _use_cpp_transport(tls_config):
+        from rayfed_amd.proxy.xfer import XferSenderService
+
+        service = XferSenderService(addresses, party, job_name, proxy_config)
+        _sender_service = service
+        _service_registry[sender_proxy_name(job_name, use_global_proxy)] = service
+        return service
     if proxy_cls is None:
         from rayfed_amd.proxy.tcp.tcp_proxy import TcpSenderProxy
 

@@ -1,0 +1,330 @@
+"""Python services over the C++ transport core (csrc/xfer_core.cpp).
+
+The native tier of the cross-silo hot path: framing, sockets, routing, the
+receive mailbox and ack round trips run in C++ threads with the GIL
+released; Python handles (de)serialization and the shm-lane consume.
+Selected automatically by ``rayfed_amd.proxy.barriers`` for plaintext jobs
+when the extension is built (``RAYFED_TRANSPORT=asyncio`` forces the Python
+transport; TLS jobs always use it).
+
+These classes implement the same *service* interface as
+``barriers.SenderProxyService`` / ``ReceiverProxyService`` (send/get_data
+returning concurrent futures, ``_get_stats``, ``stop``) — the coroutine SPI
+does not apply because nothing here runs on an event loop.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from concurrent.futures import Future, ThreadPoolExecutor
+from typing import Dict, Optional
+
+from rayfed_amd import config as fed_config
+from rayfed_amd.ops import tensor_codec
+from rayfed_amd.proxy.grpc import frames
+from rayfed_amd.runtime.object_ref import ObjectRef
+
+logger = logging.getLogger(__name__)
+
+_OBJ_MARKER = b"\x00RFOBJ"
+
+
+def _load_xfer():
+    import torch  # noqa: F401 — libc10 first
+
+    import rayfed_amd._xfer as xfer
+
+    return xfer
+
+
+def xfer_available() -> bool:
+    try:
+        _load_xfer()
+        return True
+    except ImportError:
+        return False
+
+
+def _coerce_config(proxy_config):
+    if proxy_config is not None and not isinstance(
+        proxy_config, fed_config.CrossSiloMessageConfig
+    ):
+        proxy_config = fed_config.GrpcCrossSiloMessageConfig.from_dict(proxy_config)
+    return proxy_config
+
+
+class _Retry:
+    def __init__(self, d: Optional[dict]):
+        from rayfed_amd.proxy.tcp.tcp_proxy import _RetryPolicy
+
+        self._p = _RetryPolicy(d)
+
+    @property
+    def max_attempts(self):
+        return self._p.max_attempts
+
+    @property
+    def initial_backoff(self):
+        return self._p.initial_backoff
+
+    @property
+    def max_backoff(self):
+        return self._p.max_backoff
+
+    @property
+    def multiplier(self):
+        return self._p.multiplier
+
+
+class XferSenderService:
+    """send() → concurrent Future; encode + C++ socket I/O on pool threads."""
+
+    def __init__(self, addresses: Dict, party: str, job_name: str,
+                 proxy_config=None):
+        proxy_config = _coerce_config(proxy_config)
+        self._addresses = addresses
+        self._party = party
+        self._job_name = job_name
+        self._proxy_config = proxy_config
+        self._client = _load_xfer().XferClient(job_name)
+        self._pool = ThreadPoolExecutor(max_workers=8, thread_name_prefix="xfer-send")
+        self._retry = _Retry(
+            getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
+        )
+        self._timeout_s = (
+            (proxy_config.timeout_in_ms / 1000.0)
+            if proxy_config and proxy_config.timeout_in_ms
+            else 60.0
+        )
+        self.gpu_plane = None
+        self._stats_lock = threading.Lock()
+        self.send_op_count = 0
+        self._same_host_cache: Dict[str, bool] = {}
+        from rayfed_amd.proxy.barriers import _EdgeStats
+
+        self._edges: Dict[str, object] = {}
+        self._edge_cls = _EdgeStats
+
+    # service interface -------------------------------------------------------
+    @property
+    def proxy(self):
+        return self  # config introspection parity (tests read ._proxy_config)
+
+    def _same_host(self, dest_party: str) -> bool:
+        cached = self._same_host_cache.get(dest_party)
+        if cached is not None:
+            return cached
+        import socket
+
+        host = self._addresses.get(dest_party, "").rsplit(":", 1)[0]
+        same = host in ("127.0.0.1", "localhost", "::1", socket.gethostname())
+        self._same_host_cache[dest_party] = same
+        return same
+
+    def send(self, dest_party, data, upstream_seq_id, downstream_seq_id) -> Future:
+        with self._stats_lock:
+            self.send_op_count += 1
+        return self._pool.submit(
+            self._send_blocking, dest_party, data,
+            str(upstream_seq_id), str(downstream_seq_id),
+        )
+
+    def _send_blocking(self, dest_party, data, up, down) -> bool:
+        t0 = time.perf_counter()
+        err = True
+        nbytes = 0
+        try:
+            if isinstance(data, ObjectRef):
+                data = data.result()  # producer error propagates to the future
+            from rayfed_amd.ops import shm_pool
+
+            use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
+            extras, parts = tensor_codec.encode(data, self.gpu_plane, use_shm)
+            try:
+                header = {"job": self._job_name, "up": up, "down": down}
+                defer_ack = False
+                if extras["tensors"]:
+                    wire_header = {
+                        k: v for k, v in extras.items() if k != "_releases"
+                    }
+                    header.update(wire_header)
+                    defer_ack = any("shm" in m for m in extras["tensors"])
+                    kind = frames.KIND_TENSOR
+                    body_parts = [
+                        frames.encode_frame_prefix(kind, header)
+                    ] + list(parts)
+                else:
+                    kind = frames.KIND_PICKLE
+                    body_parts = [
+                        frames.encode_frame_prefix(kind, header),
+                        parts[0],
+                    ]
+                from rayfed_amd.exceptions import FedRemoteError
+
+                if isinstance(data, FedRemoteError):
+                    # encode() pickled it as a plain object; re-tag the kind.
+                    from rayfed_amd._private import serialization
+
+                    body_parts = [
+                        frames.encode_frame_prefix(
+                            frames.KIND_ERROR,
+                            {"job": self._job_name, "up": up, "down": down},
+                        ),
+                        serialization.dumps(data),
+                    ]
+                nbytes = sum(len(p) for p in body_parts)
+                host, port = self._addresses[dest_party].rsplit(":", 1)
+                code, result = self._send_with_retry(
+                    host, int(port), up, down, body_parts, defer_ack
+                )
+            finally:
+                tensor_codec.release_parts(extras)
+            if 400 <= code < 500:
+                raise RuntimeError(
+                    f"[{code}] send to {dest_party} rejected: {result}"
+                )
+            if code >= 500:
+                raise RuntimeError(
+                    f"[{code}] send to {dest_party} failed: {result}"
+                )
+            err = False
+            return True
+        finally:
+            secs = time.perf_counter() - t0
+            with self._stats_lock:
+                edge = self._edges.setdefault(dest_party, self._edge_cls())
+                edge.record(nbytes, secs, err)
+
+    def _send_with_retry(self, host, port, up, down, parts, defer_ack):
+        deadline = time.monotonic() + self._timeout_s
+        backoff = self._retry.initial_backoff
+        attempt = 0
+        while True:
+            attempt += 1
+            try:
+                remaining = max(0.001, deadline - time.monotonic())
+                return self._client.send(
+                    host, port, up, down, parts, defer_ack, remaining
+                )
+            except RuntimeError as e:
+                now = time.monotonic()
+                if attempt >= self._retry.max_attempts or now + backoff >= deadline:
+                    raise RuntimeError(
+                        f"send failed after {attempt} attempts: {e}"
+                    ) from e
+                logger.debug("xfer send attempt %d failed (%r)", attempt, e)
+                time.sleep(backoff)
+                backoff = min(backoff * self._retry.multiplier,
+                              self._retry.max_backoff)
+
+    def _get_stats(self) -> Dict[str, object]:
+        with self._stats_lock:
+            return {
+                "send_op_count": self.send_op_count,
+                "edges": {p: e.as_dict() for p, e in self._edges.items()},
+            }
+
+    def stop(self):
+        self._pool.shutdown(wait=False, cancel_futures=True)
+        self._client.close_all()
+
+
+class XferReceiverService:
+    """get_data() → concurrent Future; C++ mailbox, Python deserialization."""
+
+    def __init__(self, listening_address: str, party: str, job_name: str,
+                 proxy_config=None):
+        proxy_config = _coerce_config(proxy_config)
+        self._party = party
+        self._job_name = job_name
+        self._proxy_config = proxy_config
+        self._allowed_list = (
+            proxy_config.serializing_allowed_list if proxy_config else None
+        )
+        port = int(listening_address.rsplit(":", 1)[1])
+        xfer = _load_xfer()
+        try:
+            self._server = xfer.XferServer(port, job_name)
+            self._server.start(self._consume_deferred)
+        except RuntimeError as e:
+            raise AssertionError(
+                f"Failed to listen on port {port}: it is in use ({e})."
+            ) from e
+        self._pool = ThreadPoolExecutor(max_workers=32, thread_name_prefix="xfer-recv")
+        self._objs: Dict[tuple, object] = {}
+        self._objs_lock = threading.Lock()
+        self.gpu_plane = None
+        self._stats_lock = threading.Lock()
+        self.receive_op_count = 0
+        self._deferred_count = 0
+
+    @property
+    def proxy(self):
+        return self
+
+    @property
+    def received_op_count(self) -> int:
+        return int(self._server.received_op_count) + self._deferred_count
+
+    # C++ calls this (with the GIL) for DEFER_ACK frames — the shm lane's
+    # consume-before-ack: decode (H2D + CRC) here, park the object, post a
+    # marker so any blocked get_data wakes.
+    def _consume_deferred(self, up: str, down: str, body: bytes) -> int:
+        try:
+            kind, header, payload = frames.decode_frame(body)
+            obj = tensor_codec.decode(
+                {k: header[k] for k in ("skel", "tensors")},
+                memoryview(payload),
+                self.gpu_plane,
+                self._allowed_list,
+            )
+        except Exception as e:  # noqa: BLE001
+            logger.warning("xfer deferred consume failed: %r", e)
+            return 500
+        with self._objs_lock:
+            self._objs[(up, down)] = obj
+            self._deferred_count += 1
+        self._server.post(up, down, _OBJ_MARKER)
+        return 200
+
+    def _take(self, up: str, down: str, body: bytes):
+        if bytes(body) == _OBJ_MARKER:
+            with self._objs_lock:
+                return self._objs.pop((up, down))
+        kind, header, payload = frames.decode_frame(body)
+        from rayfed_amd._private import serialization
+
+        if kind == frames.KIND_ERROR:
+            raise serialization.loads(payload, self._allowed_list)
+        if kind == frames.KIND_TENSOR:
+            return tensor_codec.decode(
+                header, memoryview(payload), self.gpu_plane, self._allowed_list
+            )
+        return serialization.loads(payload, self._allowed_list)
+
+    def get_data(self, src_party, upstream_seq_id, curr_seq_id) -> Future:
+        up, down = str(upstream_seq_id), str(curr_seq_id)
+        with self._stats_lock:
+            self.receive_op_count += 1
+        body = self._server.try_take(up, down)
+        if body is not None:
+            fut: Future = Future()
+            try:
+                fut.set_result(self._take(up, down, body))
+            except BaseException as e:  # noqa: BLE001
+                fut.set_exception(e)
+            return fut
+
+        def _wait():
+            b = self._server.get_data(up, down, 600.0)
+            return self._take(up, down, b)
+
+        return self._pool.submit(_wait)
+
+    def _get_stats(self) -> Dict[str, int]:
+        return {"receive_op_count": self.receive_op_count}
+
+    def stop(self):
+        self._server.stop()
+        self._pool.shutdown(wait=False, cancel_futures=True)

@@ -26,7 +26,12 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
-        )
+        ),
+        cpp_extension.CppExtension(
+            name="rayfed_amd._xfer",
+            sources=["csrc/xfer_core.cpp"],
+            extra_compile_args=["-O3", "-std=c++17", "-pthread"],
+        ),
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
 )
