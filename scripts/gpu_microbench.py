@@ -99,6 +99,58 @@ def main():
     report("D2H pinned", n, timeit(lambda: pin.copy_(dev, non_blocking=False)))
     report("H2D pinned", n, timeit(lambda: dev.copy_(pin, non_blocking=False)))
 
+    # Bidirectional DMA aggregate: is the host link full duplex?
+    n = 1 << 30
+    dev1 = torch.empty(n, dtype=torch.uint8, device="cuda")
+    dev2 = torch.empty(n, dtype=torch.uint8, device="cuda")
+    pin1 = torch.empty(n, dtype=torch.uint8, pin_memory=True)
+    pin2 = torch.empty(n, dtype=torch.uint8, pin_memory=True)
+    s_d2h = torch.cuda.Stream()
+    s_h2d = torch.cuda.Stream()
+
+    def bidir():
+        with torch.cuda.stream(s_d2h):
+            pin1.copy_(dev1, non_blocking=True)
+        with torch.cuda.stream(s_h2d):
+            dev2.copy_(pin2, non_blocking=True)
+        torch.cuda.synchronize()
+
+    report("bidirectional D2H+H2D (2 GiB moved)", 2 * n, timeit(bidir))
+
+    # Registered /dev/shm segment: does torch see it as pinned (async DMA)?
+    from rayfed_amd.ops import shm_pool
+
+    pool = shm_pool.get_send_pool()
+    seg = pool.acquire(1 << 28)
+    print(json.dumps({"op": "shm_seg", "registered": seg.registered,
+                      "torch_is_pinned": bool(seg.torch_view.is_pinned())}),
+          flush=True)
+
+    def shm_d2h():
+        seg.torch_view[: 1 << 28].copy_(dev1[: 1 << 28], non_blocking=True)
+        torch.cuda.synchronize()
+
+    report("D2H into registered shm", 1 << 28, timeit(shm_d2h))
+    pool.release(seg)
+
+    # Chunked shm pipeline e2e in-process: pack thread || unpack consume.
+    from rayfed_amd.config import GpuDataPlaneConfig as _C
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane as _P
+    pl = _P(_C())
+    t = torch.randn((1 << 30) // 4, device="cuda")  # 1 GiB f32
+
+    def chunked_e2e():
+        seg2, man, rel = pl.pack_to_shm_chunked(t)
+        man = dict(man, nbytes=t.numel() * 4, shm=seg2.name)
+        out = pl.unpack_from_shm_chunked(seg2.name, man, torch.float32,
+                                         [t.numel()])
+        rel()
+        return out
+
+    report("chunked shm pack||unpack e2e 1 GiB", 1 << 30,
+           timeit(chunked_e2e, reps=3, warm=1))
+    shm_pool.detach_all()
+
     # Full plane pipeline
     plane = GpuDataPlane(GpuDataPlaneConfig())
     for gib in [0.25, 1.0]:
