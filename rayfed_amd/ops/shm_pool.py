@@ -80,6 +80,9 @@ class Segment:
         self.registered = _maybe_register(ptr, nbytes)
         self._ptr = ptr
 
+    def view(self, off: int, nbytes: int) -> memoryview:
+        return memoryview(self.array.data)[off : off + nbytes]
+
     def close(self, unlink: bool = True):
         if self.registered:
             try:
@@ -181,10 +184,17 @@ def get_send_pool() -> ShmSegmentPool:
         return _send_pool
 
 
-def attach(name: str) -> AttachedSegment:
+def attach(name: str):
     with _lock:
         seg = _attach_cache.get(name)
         if seg is None:
+            # Same-process loopback (tests, 1-process demos): reuse the owned
+            # segment — hipHostRegister would fail on the already-registered
+            # mapping.
+            if _send_pool is not None:
+                for own in _send_pool._all:  # noqa: SLF001
+                    if own.name == name:
+                        return own
             seg = AttachedSegment(name)
             _attach_cache[name] = seg
         return seg

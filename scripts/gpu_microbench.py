@@ -133,21 +133,23 @@ def main():
     report("D2H into registered shm", 1 << 28, timeit(shm_d2h))
     pool.release(seg)
 
-    # Chunked shm pipeline e2e in-process: pack thread || unpack consume.
+    # Chunked shm pipeline e2e in-process with SEPARATE planes (distinct
+    # copy streams), approximating the two-process pipeline.
     from rayfed_amd.config import GpuDataPlaneConfig as _C
     from rayfed_amd.ops.gpu_plane import GpuDataPlane as _P
-    pl = _P(_C())
+    pl_tx = _P(_C())
+    pl_rx = _P(_C())
     t = torch.randn((1 << 30) // 4, device="cuda")  # 1 GiB f32
 
     def chunked_e2e():
-        seg2, man, rel = pl.pack_to_shm_chunked(t)
+        seg2, man, rel = pl_tx.pack_to_shm_chunked(t)
         man = dict(man, nbytes=t.numel() * 4, shm=seg2.name)
-        out = pl.unpack_from_shm_chunked(seg2.name, man, torch.float32,
-                                         [t.numel()])
+        out = pl_rx.unpack_from_shm_chunked(seg2.name, man, torch.float32,
+                                            [t.numel()])
         rel()
         return out
 
-    report("chunked shm pack||unpack e2e 1 GiB", 1 << 30,
+    report("chunked shm pack||unpack e2e 1 GiB (2 planes)", 1 << 30,
            timeit(chunked_e2e, reps=3, warm=1))
     shm_pool.detach_all()
 
