@@ -151,6 +151,14 @@ class AttachedSegment:
         self.torch_view = torch.from_numpy(self.array) if torch is not None else None
         self.registered = _maybe_register(self.array.ctypes.data, len(self.array))
         self._ptr = self.array.ctypes.data
+        if os.environ.get("RAYFED_SHM_DEBUG"):
+            import sys
+
+            print(
+                f"[shm] attached {name}: {len(self.array)>>20} MiB "
+                f"registered={self.registered}",
+                file=sys.stderr, flush=True,
+            )
 
     def view(self, off: int, nbytes: int) -> memoryview:
         return memoryview(self.array.data)[off : off + nbytes]
