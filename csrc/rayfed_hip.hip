@@ -690,6 +690,47 @@ int64_t crc32_combine_py(int64_t crc1, int64_t crc2, int64_t len2) {
                                             (uint64_t)len2);
 }
 
+// ---------------------------------------------------------------------------
+// Device IPC lane: same-node parties exchange GPU buffers directly
+// (hipIpc*/dmabuf) — no host bounce at all.  The sender packs+CRCs into a
+// dedicated hipMalloc'd staging buffer, ships the 64-byte handle; the
+// receiver opens it (cached) and D2D-copies at HBM/xGMI rate.
+// ---------------------------------------------------------------------------
+std::tuple<int64_t, py::bytes> ipc_alloc(int64_t nbytes) {
+  void* ptr = nullptr;
+  HIP_CHECK(hipMalloc(&ptr, (size_t)nbytes));
+  hipIpcMemHandle_t handle;
+  HIP_CHECK(hipIpcGetMemHandle(&handle, ptr));
+  return {reinterpret_cast<int64_t>(ptr),
+          py::bytes(reinterpret_cast<const char*>(&handle), sizeof(handle))};
+}
+
+void ipc_free(int64_t ptr) {
+  HIP_CHECK(hipFree(reinterpret_cast<void*>(ptr)));
+}
+
+int64_t ipc_open(py::bytes handle_bytes) {
+  std::string h = handle_bytes;
+  TORCH_CHECK(h.size() == sizeof(hipIpcMemHandle_t), "bad ipc handle size");
+  hipIpcMemHandle_t handle;
+  std::memcpy(&handle, h.data(), sizeof(handle));
+  void* ptr = nullptr;
+  HIP_CHECK(hipIpcOpenMemHandle(&ptr, handle, hipIpcMemLazyEnablePeerAccess));
+  return reinterpret_cast<int64_t>(ptr);
+}
+
+void ipc_close(int64_t ptr) {
+  HIP_CHECK(hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr)));
+}
+
+torch::Tensor tensor_from_ptr(int64_t ptr, int64_t nbytes, int64_t device) {
+  // Non-owning uint8 view of a raw device pointer (e.g. an opened IPC
+  // mapping); the caller guarantees lifetime until the ack.
+  auto options =
+      torch::dtype(torch::kUInt8).device(torch::kCUDA, (int)device);
+  return torch::from_blob(reinterpret_cast<void*>(ptr), {nbytes}, options);
+}
+
 // Pin an existing host mapping (e.g. a /dev/shm segment) for true-DMA
 // D2H/H2D — the same-host shm lane registers pooled segments once.
 void host_register(int64_t ptr, int64_t size) {
@@ -722,4 +763,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("crc32_combine", &crc32_combine_py, "zlib-style CRC combine");
   m.def("host_register", &host_register, "hipHostRegister an existing mapping");
   m.def("host_unregister", &host_unregister, "hipHostUnregister");
+  m.def("ipc_alloc", &ipc_alloc,
+        "hipMalloc + hipIpcGetMemHandle -> (ptr, handle bytes)");
+  m.def("ipc_free", &ipc_free);
+  m.def("ipc_open", &ipc_open, "hipIpcOpenMemHandle -> ptr");
+  m.def("ipc_close", &ipc_close);
+  m.def("tensor_from_ptr", &tensor_from_ptr,
+        "non-owning uint8 CUDA tensor over a raw device pointer");
 }

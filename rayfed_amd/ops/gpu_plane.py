@@ -60,6 +60,11 @@ class GpuDataPlane:
         self._lock = threading.Lock()
         self._pinned: List[torch.Tensor] = []
         self._dev_staging: List[torch.Tensor] = []
+        # Device-IPC lane state: pooled hipMalloc slabs we own (handle →
+        # slab) and peer handles we've opened (handle → ptr).
+        self._ipc_pool: List = []
+        self._own_ipc: dict = {}
+        self._ipc_open_cache: dict = {}
 
     # -- buffer pools ---------------------------------------------------------
     def _get_buf(self, pool: List, nbytes: int, pin: bool) -> "torch.Tensor":
@@ -141,6 +146,108 @@ class GpuDataPlane:
         view = memoryview(pinned.numpy())[:nbytes]
         return view, crc, lambda: self._put_buf(self._pinned, pinned)
 
+    # ------------------------------------------------------------------
+    # Device-IPC lane: same-NODE parties skip the host entirely.  The
+    # sender packs+CRCs into a pooled hipMalloc'd staging buffer (fused
+    # pack_crc kernel, one HBM pass), ships the 64-byte hipIpc handle in
+    # the manifest; the receiver opens it once (cached) and D2D-copies at
+    # HBM (same GPU) or xGMI (peer GPU) rate, CRC-verifying in the same
+    # fused pass.  Ack-after-consume (DEFER_ACK) licenses staging reuse,
+    # and a reused staging buffer keeps its handle — steady state has
+    # zero IPC-open cost.
+    # ------------------------------------------------------------------
+    def _ipc_get(self, nbytes: int):
+        with self._lock:
+            for i, slab in enumerate(self._ipc_pool):
+                if slab[2] >= nbytes:
+                    return self._ipc_pool.pop(i)
+        ptr, handle = self._ext.ipc_alloc(max(nbytes, 1 << 20))
+        view = self._ext.tensor_from_ptr(
+            ptr, max(nbytes, 1 << 20), self.device.index
+        )
+        slab = (ptr, bytes(handle), max(nbytes, 1 << 20), view)
+        self._own_ipc[slab[1]] = slab
+        return slab
+
+    def _ipc_put(self, slab):
+        with self._lock:
+            if len(self._ipc_pool) < self.config.staging_buffers:
+                self._ipc_pool.append(slab)
+                return
+        self._own_ipc.pop(slab[1], None)
+        self._ext.ipc_free(slab[0])
+
+    def pack_to_ipc(self, t: "torch.Tensor"):
+        """Returns (handle_bytes, man_fields, crc|None, release_fn)."""
+        t = t.detach()
+        if not t.is_contiguous():
+            t = t.contiguous()
+        wire_fp8 = (
+            self.config.wire_dtype == "fp8e4m3" and t.dtype == torch.bfloat16
+        )
+        nbytes = t.numel() if wire_fp8 else t.numel() * t.element_size()
+        slab = self._ipc_get(nbytes)
+        produced = torch.cuda.current_stream(self.device).record_event()
+        with torch.cuda.stream(self._copy_stream):
+            self._copy_stream.wait_event(produced)
+            if wire_fp8:
+                crc_out = self._ext.pack_fp8_async(t.view(-1), slab[3][:nbytes])
+            else:
+                flat = t.view(-1).view(torch.uint8)
+                crc_out = self._ext.pack_crc_async(flat, slab[3][:nbytes])
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        crc = (
+            int(crc_out[2].item()) & 0xFFFFFFFF if self.config.verify_crc else None
+        )
+        man = {"ipc": slab[1], "ipc_off": 0}
+        if wire_fp8:
+            man["wire"] = "fp8e4m3"
+        return slab[1], man, crc, lambda: self._ipc_put(slab)
+
+    def _ipc_src_view(self, handle: bytes, nbytes: int):
+        own = self._own_ipc.get(handle)
+        if own is not None:  # same-process loopback: use the local mapping
+            return own[3][:nbytes]
+        with self._lock:
+            ptr = self._ipc_open_cache.get(handle)
+            if ptr is None:
+                ptr = self._ext.ipc_open(handle)
+                self._ipc_open_cache[handle] = ptr
+        return self._ext.tensor_from_ptr(ptr, nbytes, self.device.index)
+
+    def unpack_from_ipc(self, man, dtype, shape):
+        nbytes = man["nbytes"]
+        src = self._ipc_src_view(bytes(man["ipc"]), man["ipc_off"] + nbytes)[
+            man["ipc_off"] : man["ipc_off"] + nbytes
+        ]
+        out = torch.empty(shape, dtype=dtype, device=self.device)
+        crc_expect = man.get("crc32")
+        wire_fp8 = man.get("wire") == "fp8e4m3"
+        crc_out = None
+        with torch.cuda.stream(self._copy_stream):
+            if wire_fp8:
+                if self.config.verify_crc and crc_expect is not None:
+                    crc_out = self._ext.crc32_async(src)
+                self._ext.unpack_fp8_async(src.contiguous(), out.view(-1))
+            else:
+                flat = out.view(-1).view(torch.uint8)
+                if self.config.verify_crc and crc_expect is not None:
+                    # Fused D2D copy + CRC in one pass over src.
+                    crc_out = self._ext.pack_crc_async(src.contiguous(), flat)
+                else:
+                    flat.copy_(src)
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        if crc_out is not None:
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
+            if crc != crc_expect:
+                raise ValueError(
+                    f"GPU tensor CRC mismatch (ipc lane): expected "
+                    f"{crc_expect:#x}, got {crc:#x}"
+                )
+        return out
+
     # Chunk-pipelined shm pushes: the first SHM_HDR bytes of the segment are
     # a progress page — u64 chunks-ready counter at offset 0, then one u32
     # CRC per chunk at offset 8 — written by the sender as each chunk's D2H
@@ -166,6 +273,14 @@ class GpuDataPlane:
         produced = torch.cuda.current_stream(self.device).record_event()
 
         def _drive():
+            import os as _os
+            import sys as _sys
+            import time as _t
+
+            dbg = _os.environ.get("RAYFED_SHM_DEBUG")
+            if dbg:
+                print(f"[shm-tx] drive start {_t.monotonic():.4f}",
+                      file=_sys.stderr, flush=True)
             try:
                 torch.cuda.set_device(self.device)
                 events = []
@@ -200,6 +315,12 @@ class GpuDataPlane:
                         struct.pack_into("<I", seg.array, 8 + 4 * i, crc)
                     # Publish chunk i (x86: aligned 8-byte store is atomic).
                     struct.pack_into("<Q", seg.array, 0, i + 1)
+                if dbg:
+                    import sys as _sys
+                    import time as _t
+
+                    print(f"[shm-tx] all published {_t.monotonic():.4f}",
+                          file=_sys.stderr, flush=True)
             except Exception:  # noqa: BLE001
                 logger.exception("chunked shm pack failed")
                 struct.pack_into("<q", seg.array, 0, -1)  # poison
@@ -233,6 +354,13 @@ class GpuDataPlane:
         chunk = man["chunked"]
         hdr = man["hdr"]
         n_chunks = (nbytes + chunk - 1) // chunk
+        import os as _os
+        import sys as _sys
+
+        dbg = _os.environ.get("RAYFED_SHM_DEBUG")
+        if dbg:
+            print(f"[shm-rx] unpack start {_time.monotonic():.4f}",
+                  file=_sys.stderr, flush=True)
         out = torch.empty(shape, dtype=dtype, device=self.device)
         flat = out.view(-1).view(torch.uint8)
         crc_outs = []
@@ -277,6 +405,9 @@ class GpuDataPlane:
                         f"GPU tensor CRC mismatch on chunk {i}: "
                         f"expected {want:#x}, got {got:#x}"
                     )
+        if dbg:
+            print(f"[shm-rx] unpack done {_time.monotonic():.4f}",
+                  file=_sys.stderr, flush=True)
         return out
 
     def pack_to_shm(self, t: "torch.Tensor"):

@@ -150,10 +150,22 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
             from rayfed_amd.ops import shm_pool
 
             if gpu_plane is not None and t.device.type == "cuda":
+                import os as _os
+
                 wire_fp8 = (
                     gpu_plane.config.wire_dtype == "fp8e4m3"
                     and t.dtype == torch.bfloat16
                 )
+                if _os.environ.get("RAYFED_IPC", "1") != "0":
+                    # Device-IPC lane: no host bounce at all (same node).
+                    _, man_fields, crc, release = gpu_plane.pack_to_ipc(t)
+                    man.update(man_fields)
+                    man["crc32"] = crc
+                    if wire_fp8:
+                        man["nbytes"] = t.numel()
+                    releases.append(release)
+                    manifests.append(man)
+                    continue
                 if not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
                     # Chunk pipeline: D2H publishes progress as it goes; the
                     # frame leaves before the DMA finishes and the receiver's
@@ -283,6 +295,17 @@ def decode(
     tensors: List[Any] = []
     for man in extras["tensors"]:
         nbytes = man["nbytes"]
+        if "ipc" in man:
+            if gpu_plane is None:
+                raise RuntimeError(
+                    "received a device-IPC tensor but no GPU data plane is "
+                    "attached (set RAYFED_IPC=0 on the sender for CPU peers)"
+                )
+            tensors.append(
+                gpu_plane.unpack_from_ipc(man, _STR_TO_DTYPE[man["dtype"]],
+                                          man["shape"])
+            )
+            continue
         if "chunked" in man:
             tensors.append(_decode_chunked(man, gpu_plane))
             continue

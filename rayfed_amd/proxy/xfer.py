@@ -149,7 +149,9 @@ class XferSenderService:
                         k: v for k, v in extras.items() if k != "_releases"
                     }
                     header.update(wire_header)
-                    defer_ack = any("shm" in m for m in extras["tensors"])
+                    defer_ack = any(
+                        "shm" in m or "ipc" in m for m in extras["tensors"]
+                    )
                     kind = frames.KIND_TENSOR
                     body_parts = [
                         frames.encode_frame_prefix(kind, header)

@@ -199,11 +199,45 @@ def test_crc_bandwidth_floor(ext):
 
 
 @needs_gpu
-def test_plane_shm_lane_gpu_roundtrip(plane):
+def test_plane_ipc_lane_roundtrip(plane):
+    """GPU tensor through the device-IPC lane (same-process decode uses the
+    owned mapping; cross-process covered by smoke/bench and ipc_probe)."""
+    from rayfed_amd.ops import tensor_codec
+
+    t = torch.randn(1 << 20, device="cuda")
+    extras, parts = tensor_codec.encode(t, plane, shm=True)
+    man = extras["tensors"][0]
+    assert "ipc" in man and man["crc32"] is not None
+    payload = b"".join(bytes(p) for p in parts)
+    out = tensor_codec.decode(extras, memoryview(payload), plane, None)
+    tensor_codec.release_parts(extras)
+    assert out.is_cuda and torch.equal(out, t)
+
+
+@needs_gpu
+def test_plane_ipc_lane_fp8_wire():
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+
+    p8 = GpuDataPlane(GpuDataPlaneConfig(wire_dtype="fp8e4m3"))
+    t = (torch.randn(65536, device="cuda") * 2).to(torch.bfloat16)
+    extras, parts = tensor_codec.encode(t, p8, shm=True)
+    man = extras["tensors"][0]
+    assert "ipc" in man and man.get("wire") == "fp8e4m3"
+    out = tensor_codec.decode(extras, memoryview(b"".join(bytes(p) for p in parts)), p8, None)
+    tensor_codec.release_parts(extras)
+    ref = t.to(torch.float8_e4m3fn).to(torch.bfloat16)
+    assert torch.equal(out, ref)
+
+
+@needs_gpu
+def test_plane_shm_lane_gpu_roundtrip(plane, monkeypatch):
     """GPU tensor through the same-host shm lane: D2H DMA into a registered
     segment, H2D DMA out, CRC verified on device."""
     from rayfed_amd.ops import shm_pool, tensor_codec
 
+    monkeypatch.setenv("RAYFED_IPC", "0")
     t = torch.randn(1 << 20, device="cuda")
     extras, parts = tensor_codec.encode(t, plane, shm=True)
     man = extras["tensors"][0]
@@ -233,11 +267,12 @@ def test_shm_segment_registration(ext):
 
 
 @needs_gpu
-def test_chunked_shm_pipeline_roundtrip(plane):
+def test_chunked_shm_pipeline_roundtrip(plane, monkeypatch):
     """Chunk-pipelined shm push: sender publishes progress per chunk, the
     receiver overlaps H2D; per-chunk CRCs verified on device."""
     from rayfed_amd.ops import shm_pool, tensor_codec
 
+    monkeypatch.setenv("RAYFED_IPC", "0")
     n = (plane.config.chunk_bytes * 3) // 4 * 4  # ~3 chunks of f32
     t = torch.randn(n // 4, device="cuda")
     extras, parts = tensor_codec.encode(t, plane, shm=True)
@@ -251,11 +286,13 @@ def test_chunked_shm_pipeline_roundtrip(plane):
 
 
 @needs_gpu
-def test_chunked_shm_crc_tamper(plane):
+def test_chunked_shm_crc_tamper(plane, monkeypatch):
     """Flipping a byte in the segment after publish is caught per chunk."""
     import struct
 
     from rayfed_amd.ops import shm_pool, tensor_codec
+
+    monkeypatch.setenv("RAYFED_IPC", "0")
 
     n_el = plane.config.chunk_bytes // 2  # 2 chunks of f32
     t = torch.randn(n_el, device="cuda")
