@@ -221,7 +221,8 @@ def test_plane_ipc_lane_fp8_wire():
     from rayfed_amd.ops.gpu_plane import GpuDataPlane
 
     p8 = GpuDataPlane(GpuDataPlaneConfig(wire_dtype="fp8e4m3"))
-    t = (torch.randn(65536, device="cuda") * 2).to(torch.bfloat16)
+    # >= 1 MiB so the same-host lane engages (SHM_MIN_BYTES threshold).
+    t = (torch.randn(1 << 20, device="cuda") * 2).to(torch.bfloat16)
     extras, parts = tensor_codec.encode(t, p8, shm=True)
     man = extras["tensors"][0]
     assert "ipc" in man and man.get("wire") == "fp8e4m3"
