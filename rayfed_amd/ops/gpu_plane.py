@@ -156,29 +156,37 @@ class GpuDataPlane:
     # and a reused staging buffer keeps its handle — steady state has
     # zero IPC-open cost.
     # ------------------------------------------------------------------
+    # Staging is split into <= IPC_SLAB_BYTES slabs, each with its own
+    # handle: hipIpcOpenMemHandle was measured to HANG importing a 2 GiB
+    # dmabuf on this driver (1 MiB/256 MiB/4 GiB fine) — bounded slabs
+    # sidestep the size-dependent behavior and pool perfectly (every slab
+    # identical), so steady state re-opens nothing.
+    IPC_SLAB_BYTES = 1 << 30
+
     def _ipc_get(self, nbytes: int):
+        nbytes = min(nbytes, self.IPC_SLAB_BYTES)
         with self._lock:
-            for i, slab in enumerate(self._ipc_pool):
-                if slab[2] >= nbytes:
-                    return self._ipc_pool.pop(i)
-        ptr, handle = self._ext.ipc_alloc(max(nbytes, 1 << 20))
+            if self._ipc_pool:
+                return self._ipc_pool.pop()
+        ptr, handle = self._ext.ipc_alloc(self.IPC_SLAB_BYTES)
         view = self._ext.tensor_from_ptr(
-            ptr, max(nbytes, 1 << 20), self.device.index
+            ptr, self.IPC_SLAB_BYTES, self.device.index
         )
-        slab = (ptr, bytes(handle), max(nbytes, 1 << 20), view)
+        slab = (ptr, bytes(handle), self.IPC_SLAB_BYTES, view)
         self._own_ipc[slab[1]] = slab
         return slab
 
     def _ipc_put(self, slab):
         with self._lock:
-            if len(self._ipc_pool) < self.config.staging_buffers:
+            if len(self._ipc_pool) < max(self.config.staging_buffers, 20):
                 self._ipc_pool.append(slab)
                 return
         self._own_ipc.pop(slab[1], None)
         self._ext.ipc_free(slab[0])
 
     def pack_to_ipc(self, t: "torch.Tensor"):
-        """Returns (handle_bytes, man_fields, crc|None, release_fn)."""
+        """Pack (+CRC, fused) into one or more pooled IPC slabs.
+        Returns (handles, man_fields, crc_list|None, release_fn)."""
         t = t.detach()
         if not t.is_contiguous():
             t = t.contiguous()
@@ -186,24 +194,44 @@ class GpuDataPlane:
             self.config.wire_dtype == "fp8e4m3" and t.dtype == torch.bfloat16
         )
         nbytes = t.numel() if wire_fp8 else t.numel() * t.element_size()
-        slab = self._ipc_get(nbytes)
+        S = self.IPC_SLAB_BYTES
+        n_slabs = max(1, (nbytes + S - 1) // S)
+        slabs = [self._ipc_get(min(S, nbytes - i * S)) for i in range(n_slabs)]
         produced = torch.cuda.current_stream(self.device).record_event()
+        crc_outs = []
         with torch.cuda.stream(self._copy_stream):
             self._copy_stream.wait_event(produced)
-            if wire_fp8:
-                crc_out = self._ext.pack_fp8_async(t.view(-1), slab[3][:nbytes])
-            else:
-                flat = t.view(-1).view(torch.uint8)
-                crc_out = self._ext.pack_crc_async(flat, slab[3][:nbytes])
+            for i, slab in enumerate(slabs):
+                lo, hi = i * S, min((i + 1) * S, nbytes)
+                if wire_fp8:
+                    crc_outs.append(
+                        self._ext.pack_fp8_async(
+                            t.view(-1)[lo:hi], slab[3][: hi - lo]
+                        )
+                    )
+                else:
+                    flat = t.view(-1).view(torch.uint8)
+                    crc_outs.append(
+                        self._ext.pack_crc_async(flat[lo:hi], slab[3][: hi - lo])
+                    )
             done = self._copy_stream.record_event()
         done.synchronize()
-        crc = (
-            int(crc_out[2].item()) & 0xFFFFFFFF if self.config.verify_crc else None
+        crcs = (
+            [int(c[2].item()) & 0xFFFFFFFF for c in crc_outs]
+            if self.config.verify_crc
+            else None
         )
-        man = {"ipc": slab[1], "ipc_off": 0}
+        man = {"ipc_slabs": [s[1] for s in slabs], "slab_bytes": S}
+        if crcs is not None:
+            man["ipc_crcs"] = crcs
         if wire_fp8:
             man["wire"] = "fp8e4m3"
-        return slab[1], man, crc, lambda: self._ipc_put(slab)
+
+        def release():
+            for s in slabs:
+                self._ipc_put(s)
+
+        return [s[1] for s in slabs], man, crcs, release
 
     def _ipc_src_view(self, handle: bytes, nbytes: int):
         own = self._own_ipc.get(handle)
@@ -218,33 +246,35 @@ class GpuDataPlane:
 
     def unpack_from_ipc(self, man, dtype, shape):
         nbytes = man["nbytes"]
-        src = self._ipc_src_view(bytes(man["ipc"]), man["ipc_off"] + nbytes)[
-            man["ipc_off"] : man["ipc_off"] + nbytes
-        ]
-        out = torch.empty(shape, dtype=dtype, device=self.device)
-        crc_expect = man.get("crc32")
+        S = man["slab_bytes"]
+        handles = [bytes(h) for h in man["ipc_slabs"]]
+        crcs = man.get("ipc_crcs")
         wire_fp8 = man.get("wire") == "fp8e4m3"
-        crc_out = None
+        out = torch.empty(shape, dtype=dtype, device=self.device)
+        crc_outs = []
         with torch.cuda.stream(self._copy_stream):
-            if wire_fp8:
-                if self.config.verify_crc and crc_expect is not None:
-                    crc_out = self._ext.crc32_async(src)
-                self._ext.unpack_fp8_async(src.contiguous(), out.view(-1))
-            else:
-                flat = out.view(-1).view(torch.uint8)
-                if self.config.verify_crc and crc_expect is not None:
-                    # Fused D2D copy + CRC in one pass over src.
-                    crc_out = self._ext.pack_crc_async(src.contiguous(), flat)
+            for i, h in enumerate(handles):
+                lo, hi = i * S, min((i + 1) * S, nbytes)
+                src = self._ipc_src_view(h, hi - lo)[: hi - lo]
+                if wire_fp8:
+                    if self.config.verify_crc and crcs is not None:
+                        crc_outs.append(self._ext.crc32_async(src))
+                    self._ext.unpack_fp8_async(src, out.view(-1)[lo:hi])
                 else:
-                    flat.copy_(src)
+                    flat = out.view(-1).view(torch.uint8)
+                    if self.config.verify_crc and crcs is not None:
+                        # Fused D2D copy + CRC in one pass over src.
+                        crc_outs.append(self._ext.pack_crc_async(src, flat[lo:hi]))
+                    else:
+                        flat[lo:hi].copy_(src)
             done = self._copy_stream.record_event()
         done.synchronize()
-        if crc_out is not None:
-            crc = int(crc_out[2].item()) & 0xFFFFFFFF
-            if crc != crc_expect:
+        for i, c in enumerate(crc_outs):
+            got = int(c[2].item()) & 0xFFFFFFFF
+            if got != crcs[i]:
                 raise ValueError(
-                    f"GPU tensor CRC mismatch (ipc lane): expected "
-                    f"{crc_expect:#x}, got {crc:#x}"
+                    f"GPU tensor CRC mismatch (ipc lane, slab {i}): expected "
+                    f"{crcs[i]:#x}, got {got:#x}"
                 )
         return out
 

@@ -158,9 +158,8 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
                 )
                 if _os.environ.get("RAYFED_IPC", "1") != "0":
                     # Device-IPC lane: no host bounce at all (same node).
-                    _, man_fields, crc, release = gpu_plane.pack_to_ipc(t)
-                    man.update(man_fields)
-                    man["crc32"] = crc
+                    _, man_fields, _crcs, release = gpu_plane.pack_to_ipc(t)
+                    man.update(man_fields)  # ipc_slabs/slab_bytes/ipc_crcs/wire
                     if wire_fp8:
                         man["nbytes"] = t.numel()
                     releases.append(release)
@@ -295,7 +294,7 @@ def decode(
     tensors: List[Any] = []
     for man in extras["tensors"]:
         nbytes = man["nbytes"]
-        if "ipc" in man:
+        if "ipc_slabs" in man:
             if gpu_plane is None:
                 raise RuntimeError(
                     "received a device-IPC tensor but no GPU data plane is "

@@ -150,7 +150,7 @@ class XferSenderService:
                     }
                     header.update(wire_header)
                     defer_ack = any(
-                        "shm" in m or "ipc" in m for m in extras["tensors"]
+                        "shm" in m or "ipc_slabs" in m for m in extras["tensors"]
                     )
                     kind = frames.KIND_TENSOR
                     body_parts = [

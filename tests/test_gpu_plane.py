@@ -207,7 +207,7 @@ def test_plane_ipc_lane_roundtrip(plane):
     t = torch.randn(1 << 20, device="cuda")
     extras, parts = tensor_codec.encode(t, plane, shm=True)
     man = extras["tensors"][0]
-    assert "ipc" in man and man["crc32"] is not None
+    assert "ipc_slabs" in man and man.get("ipc_crcs")
     payload = b"".join(bytes(p) for p in parts)
     out = tensor_codec.decode(extras, memoryview(payload), plane, None)
     tensor_codec.release_parts(extras)
@@ -225,7 +225,7 @@ def test_plane_ipc_lane_fp8_wire():
     t = (torch.randn(1 << 20, device="cuda") * 2).to(torch.bfloat16)
     extras, parts = tensor_codec.encode(t, p8, shm=True)
     man = extras["tensors"][0]
-    assert "ipc" in man and man.get("wire") == "fp8e4m3"
+    assert "ipc_slabs" in man and man.get("wire") == "fp8e4m3"
     out = tensor_codec.decode(extras, memoryview(b"".join(bytes(p) for p in parts)), p8, None)
     tensor_codec.release_parts(extras)
     ref = t.to(torch.float8_e4m3fn).to(torch.bfloat16)
