@@ -336,3 +336,22 @@ def test_fedavg_reduce_mfma_matches_reference(ext):
         ref = sum(wi * x.float() for wi, x in zip(wq, ins)).to(torch.bfloat16)
         diff = (out.float() - ref.float()).abs().max().item()
         assert diff <= 0.06, f"k={k}: max diff {diff}"
+
+
+@needs_gpu
+def test_ipc_lane_multi_slab_roundtrip(plane):
+    """Tensors above IPC_SLAB_BYTES split across slabs (per-slab handles and
+    CRCs); verifies slab-boundary arithmetic for the plain path."""
+    from rayfed_amd.ops import tensor_codec
+
+    n = plane.IPC_SLAB_BYTES + (plane.IPC_SLAB_BYTES // 4)  # 1.25 GiB
+    t = torch.empty(n, dtype=torch.uint8, device="cuda")
+    t[: 1 << 22].random_()
+    t[-(1 << 22):].random_()
+    extras, parts = tensor_codec.encode(t, plane, shm=True)
+    man = extras["tensors"][0]
+    assert len(man["ipc_slabs"]) == 2 and len(man["ipc_crcs"]) == 2
+    payload = b"".join(bytes(p) for p in parts)
+    out = tensor_codec.decode(extras, memoryview(payload), plane, None)
+    tensor_codec.release_parts(extras)
+    assert torch.equal(out, t)
