@@ -137,12 +137,22 @@ class XferSenderService:
         try:
             if isinstance(data, ObjectRef):
                 data = data.result()  # producer error propagates to the future
+            from rayfed_amd.exceptions import FedRemoteError
             from rayfed_amd.ops import shm_pool
 
-            use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
-            extras, parts = tensor_codec.encode(data, self.gpu_plane, use_shm)
-            try:
-                header = {"job": self._job_name, "up": up, "down": down}
+            header = {"job": self._job_name, "up": up, "down": down}
+            if isinstance(data, FedRemoteError):
+                from rayfed_amd._private import serialization
+
+                extras = {"tensors": []}
+                body_parts = [
+                    frames.encode_frame_prefix(frames.KIND_ERROR, header),
+                    serialization.dumps(data),
+                ]
+                defer_ack = False
+            else:
+                use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
+                extras, parts = tensor_codec.encode(data, self.gpu_plane, use_shm)
                 defer_ack = False
                 if extras["tensors"]:
                     wire_header = {
@@ -152,29 +162,15 @@ class XferSenderService:
                     defer_ack = any(
                         "shm" in m or "ipc_slabs" in m for m in extras["tensors"]
                     )
-                    kind = frames.KIND_TENSOR
                     body_parts = [
-                        frames.encode_frame_prefix(kind, header)
+                        frames.encode_frame_prefix(frames.KIND_TENSOR, header)
                     ] + list(parts)
                 else:
-                    kind = frames.KIND_PICKLE
                     body_parts = [
-                        frames.encode_frame_prefix(kind, header),
+                        frames.encode_frame_prefix(frames.KIND_PICKLE, header),
                         parts[0],
                     ]
-                from rayfed_amd.exceptions import FedRemoteError
-
-                if isinstance(data, FedRemoteError):
-                    # encode() pickled it as a plain object; re-tag the kind.
-                    from rayfed_amd._private import serialization
-
-                    body_parts = [
-                        frames.encode_frame_prefix(
-                            frames.KIND_ERROR,
-                            {"job": self._job_name, "up": up, "down": down},
-                        ),
-                        serialization.dumps(data),
-                    ]
+            try:
                 nbytes = sum(len(p) for p in body_parts)
                 host, port = self._addresses[dest_party].rsplit(":", 1)
                 code, result = self._send_with_retry(
