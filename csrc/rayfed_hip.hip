@@ -71,15 +71,18 @@ __device__ __forceinline__ uint32_t crc_step8(const uint32_t* t, uint32_t crc,
 struct CrcShared {
   uint32_t tables[8 * 256];
   uint32_t pows[kMaxPow * 32];
+  uint32_t halfmat[32];  // shift matrix for kSubLen/2 zero bytes
 };
 
 __device__ __forceinline__ void load_crc_shared(CrcShared& sh,
                                                 const uint32_t* g_tables,
-                                                const uint32_t* g_pows) {
+                                                const uint32_t* g_pows,
+                                                const uint32_t* g_halfmat) {
   for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x)
     sh.tables[i] = g_tables[i];
   for (int i = threadIdx.x; i < kMaxPow * 32; i += blockDim.x)
     sh.pows[i] = g_pows[i];
+  if (threadIdx.x < 32) sh.halfmat[threadIdx.x] = g_halfmat[threadIdx.x];
   __syncthreads();
 }
 
@@ -112,10 +115,11 @@ __global__ void crc32_kernel(const uint8_t* __restrict__ data,
                              unsigned long long n, uint32_t T,
                              const uint32_t* __restrict__ g_tables,
                              const uint32_t* __restrict__ g_pows,
+                             const uint32_t* __restrict__ g_halfmat,
                              uint32_t* __restrict__ out) {
   __shared__ CrcShared sh;
   __shared__ uint32_t lds_scratch[kBlock / kWave];
-  load_crc_shared(sh, g_tables, g_pows);
+  load_crc_shared(sh, g_tables, g_pows, g_halfmat);
 
   const uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
   uint32_t contrib_full = 0, contrib_tail = 0;
@@ -126,7 +130,28 @@ __global__ void crc32_kernel(const uint8_t* __restrict__ data,
     const uint8_t* p = data + off;
     uint8_t* q = kPack ? dst + off : nullptr;
     uint32_t i = 0;
-    if ((((uintptr_t)p) & 15u) == 0) {
+    if ((((uintptr_t)p) & 15u) == 0 && len == kSubLen) {
+      // Full slice: TWO independent CRC streams (halves of the slice) to
+      // break the serial table-lookup dependency chain — the kernel is
+      // latency-bound otherwise (PMC: 85 % SQ_WAIT).  Streams combine with
+      // the half-length shift matrix at the end.
+      constexpr uint32_t H = kSubLen / 2;
+      uint32_t crc_b = 0;
+      for (; i + 16 <= H; i += 16) {
+        const uint4 va = *reinterpret_cast<const uint4*>(p + i);
+        const uint4 vb = *reinterpret_cast<const uint4*>(p + H + i);
+        if (kPack) {
+          *reinterpret_cast<uint4*>(q + i) = va;
+          *reinterpret_cast<uint4*>(q + H + i) = vb;
+        }
+        crc = crc_step8(sh.tables, crc, va.x, va.y);
+        crc_b = crc_step8(sh.tables, crc_b, vb.x, vb.y);
+        crc = crc_step8(sh.tables, crc, va.z, va.w);
+        crc_b = crc_step8(sh.tables, crc_b, vb.z, vb.w);
+      }
+      crc = gf2_apply(sh.halfmat, crc) ^ crc_b;
+      i = kSubLen;
+    } else if ((((uintptr_t)p) & 15u) == 0) {
       for (; i + 16 <= len; i += 16) {
         const uint4 v = *reinterpret_cast<const uint4*>(p + i);
         if (kPack) *reinterpret_cast<uint4*>(q + i) = v;
@@ -212,10 +237,11 @@ __global__ void pack_fp8_kernel(const uint16_t* __restrict__ src,
                                 unsigned long long n_elems, uint32_t T,
                                 const uint32_t* __restrict__ g_tables,
                                 const uint32_t* __restrict__ g_pows,
+                                const uint32_t* __restrict__ g_halfmat,
                                 uint32_t* __restrict__ out) {
   __shared__ CrcShared sh;
   __shared__ uint32_t lds_scratch[kBlock / kWave];
-  load_crc_shared(sh, g_tables, g_pows);
+  load_crc_shared(sh, g_tables, g_pows, g_halfmat);
 
   const uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
   uint32_t contrib_full = 0, contrib_tail = 0;
@@ -458,8 +484,9 @@ __global__ void masked_add_kernel(T* __restrict__ dst,
 // host-side state: tables + power matrices on device (uploaded once)
 // ---------------------------------------------------------------------------
 struct DeviceConsts {
-  uint32_t* tables = nullptr;  // 8*256
-  uint32_t* pows = nullptr;    // kMaxPow*32
+  uint32_t* tables = nullptr;   // 8*256
+  uint32_t* pows = nullptr;     // kMaxPow*32
+  uint32_t* halfmat = nullptr;  // 32 (shift by kSubLen/2)
 };
 
 DeviceConsts& get_device_consts() {
@@ -476,11 +503,16 @@ DeviceConsts& get_device_consts() {
     for (int j = 1; j < kMaxPow; ++j) {
       rayfed_crc::gf2_square(&pows[j * 32], &pows[(j - 1) * 32]);
     }
+    uint32_t halfmat[32];
+    rayfed_crc::shift_matrix(halfmat, kSubLen / 2);
     HIP_CHECK(hipMalloc(&consts.tables, tables.size() * 4));
     HIP_CHECK(hipMalloc(&consts.pows, pows.size() * 4));
+    HIP_CHECK(hipMalloc(&consts.halfmat, sizeof(halfmat)));
     HIP_CHECK(hipMemcpy(consts.tables, tables.data(), tables.size() * 4,
                         hipMemcpyHostToDevice));
     HIP_CHECK(hipMemcpy(consts.pows, pows.data(), pows.size() * 4,
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(consts.halfmat, halfmat, sizeof(halfmat),
                         hipMemcpyHostToDevice));
   });
   return consts;
@@ -516,7 +548,7 @@ torch::Tensor crc32_async(torch::Tensor bytes) {
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(crc32_kernel<false>, dim3(blocks), dim3(kBlock), 0,
                      stream, bytes.data_ptr<uint8_t>(), nullptr, n, T,
-                     consts.tables, consts.pows,
+                     consts.tables, consts.pows, consts.halfmat,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
   hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
@@ -546,7 +578,7 @@ torch::Tensor pack_crc_async(torch::Tensor src, torch::Tensor dst) {
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(crc32_kernel<true>, dim3(blocks), dim3(kBlock), 0,
                      stream, src.data_ptr<uint8_t>(), dst.data_ptr<uint8_t>(),
-                     n, T, consts.tables, consts.pows,
+                     n, T, consts.tables, consts.pows, consts.halfmat,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
   hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
@@ -572,6 +604,7 @@ torch::Tensor pack_fp8_async(torch::Tensor src, torch::Tensor dst) {
   hipLaunchKernelGGL(pack_fp8_kernel, dim3(blocks), dim3(kBlock), 0, stream,
                      reinterpret_cast<const uint16_t*>(src.data_ptr()),
                      dst.data_ptr<uint8_t>(), n, T, consts.tables, consts.pows,
+                     consts.halfmat,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()));
   hipLaunchKernelGGL(crc_finalize_kernel, dim3(1), dim3(kWave), 0, stream,
                      reinterpret_cast<uint32_t*>(out.data_ptr<int32_t>()),
