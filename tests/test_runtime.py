@@ -133,3 +133,20 @@ def test_message_queue_is_event_driven():
     dt = time.perf_counter() - t0
     mq.stop()
     assert dt < 0.05, f"queue handling took {dt*1e3:.1f} ms — not event-driven"
+
+
+def test_shutdown_flag_acquired_once():
+    """Concurrent failure paths trigger exactly one shutdown
+    (parity: reference global_context.py:70-87 / cleanup shutdown-once)."""
+    from rayfed_amd._private.global_context import GlobalContext
+
+    ctx = GlobalContext("j", "alice")
+    try:
+        import concurrent.futures as cf
+
+        with cf.ThreadPoolExecutor(8) as pool:
+            results = list(pool.map(lambda _: ctx.acquire_shutdown_flag(), range(32)))
+        assert sum(results) == 1
+    finally:
+        ctx.get_cleanup_manager().stop(wait_for_sending=False)
+        ctx.get_executor().shutdown(wait=False)

@@ -11,9 +11,27 @@ from rayfed_amd._private.global_context import (
 from tests._util import make_addresses
 
 
-@pytest.fixture(params=["tcp", "grpc"])
-def party_env(request):
-    if request.param == "tcp":
+@pytest.fixture(scope="session")
+def _tls_conf(tmp_path_factory):
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    from tool.generate_tls_certs import generate
+
+    conf = generate(str(tmp_path_factory.mktemp("tp-certs")))
+    conf["target_name_override"] = "localhost"
+    return conf
+
+
+@pytest.fixture(params=["tcp", "grpc", "tcp-tls"])
+def party_env(request, _tls_conf):
+    tls = None
+    if request.param == "tcp-tls":
+        # Transport-level TLS with mutual auth (coverage parity:
+        # reference test_transport_proxy_tls.py).
+        tls = _tls_conf
+    if request.param.startswith("tcp"):
         from rayfed_amd.proxy.tcp.tcp_proxy import TcpReceiverProxy, TcpSenderProxy
 
         sender_cls, receiver_cls = TcpSenderProxy, TcpReceiverProxy
@@ -27,10 +45,12 @@ def party_env(request):
     addrs = make_addresses(["alice"])
     init_global_context(current_party="alice", job_name="test_job")
     receiver = barriers.start_receiver_proxy(
-        addrs, "alice", job_name="test_job", proxy_cls=receiver_cls, proxy_config=None
+        addrs, "alice", job_name="test_job", tls_config=tls,
+        proxy_cls=receiver_cls, proxy_config=None,
     )
     sender = barriers.start_sender_proxy(
-        addrs, "alice", job_name="test_job", proxy_cls=sender_cls, proxy_config=None
+        addrs, "alice", job_name="test_job", tls_config=tls,
+        proxy_cls=sender_cls, proxy_config=None,
     )
     yield request.param, addrs, sender, receiver
     clear_global_context()
@@ -72,6 +92,8 @@ def test_wrong_job_name_rejected_with_417(party_env):
         {"job": "SOME_OTHER_JOB", "up": "1", "down": "1"},
         b"payload",
     )
+    if kind == "tcp-tls":
+        pytest.skip("raw-plaintext probe not applicable to the TLS listener")
     if kind == "grpc":
         import grpc
 
