@@ -66,6 +66,12 @@ class GpuDataPlane:
         self._own_ipc: dict = {}
         self._ipc_open_cache: dict = {}
 
+    def _bind_device(self):
+        """hipSetDevice is per-THREAD: transport pool threads and the C++
+        consume thread default to device 0, so every entrypoint binds the
+        plane's device first (multi-GPU ranks pin one device per process)."""
+        torch.cuda.set_device(self.device)
+
     # -- buffer pools ---------------------------------------------------------
     def _get_buf(self, pool: List, nbytes: int, pin: bool) -> "torch.Tensor":
         with self._lock:
@@ -94,6 +100,7 @@ class GpuDataPlane:
         kernel and the D2H DMA run on separate streams off the producing
         stream's event, overlapping each other.
         """
+        self._bind_device()
         t = t.detach()
         if not t.is_contiguous():
             t = t.contiguous()
@@ -187,6 +194,7 @@ class GpuDataPlane:
     def pack_to_ipc(self, t: "torch.Tensor"):
         """Pack (+CRC, fused) into one or more pooled IPC slabs.
         Returns (handles, man_fields, crc_list|None, release_fn)."""
+        self._bind_device()
         t = t.detach()
         if not t.is_contiguous():
             t = t.contiguous()
@@ -245,6 +253,7 @@ class GpuDataPlane:
         return self._ext.tensor_from_ptr(ptr, nbytes, self.device.index)
 
     def unpack_from_ipc(self, man, dtype, shape):
+        self._bind_device()
         nbytes = man["nbytes"]
         S = man["slab_bytes"]
         handles = [bytes(h) for h in man["ipc_slabs"]]
@@ -289,6 +298,7 @@ class GpuDataPlane:
         IMMEDIATELY with (segment, manifest_fields, release_fn); a worker
         thread advances the progress counter as chunk DMAs complete.
         Only for plain-wire contiguous tensors above ~2 chunks."""
+        self._bind_device()
         import struct
 
         from rayfed_amd.ops import shm_pool
@@ -374,6 +384,7 @@ class GpuDataPlane:
         """Receiver side of the chunk pipeline: H2D each chunk as soon as
         the sender publishes it, CRC-verify on device, overlap everything on
         the copy stream."""
+        self._bind_device()
         import struct
         import time as _time
 
@@ -444,6 +455,7 @@ class GpuDataPlane:
         """Like :meth:`pack_to_host` but the destination is a pooled
         /dev/shm segment (hipHostRegister-ed once) so a same-host peer can
         H2D straight out of it.  Returns (segment, crc|None, release_fn)."""
+        self._bind_device()
         from rayfed_amd.ops import shm_pool
 
         t = t.detach()
@@ -510,6 +522,7 @@ class GpuDataPlane:
         wire_dtype: Optional[str] = None,
         src_tensor: Optional["torch.Tensor"] = None,
     ) -> "torch.Tensor":
+        self._bind_device()
         nbytes = len(raw)
         out = torch.empty(shape, dtype=dtype, device=self.device)
         if nbytes == 0:
