@@ -207,23 +207,39 @@ class GpuDataPlane:
         slabs = [self._ipc_get(min(S, nbytes - i * S)) for i in range(n_slabs)]
         produced = torch.cuda.current_stream(self.device).record_event()
         crc_outs = []
+        # Copy and CRC run on SEPARATE streams, both reading the source: the
+        # CRC kernel tops out ≈0.6 TB/s (LDS-lookup bound) while a plain D2D
+        # copy runs ≈2.8 TB/s, so fusing them serializes the copy behind the
+        # CRC — overlapped, the phase costs max(copy, crc), ~2× faster.
         with torch.cuda.stream(self._copy_stream):
             self._copy_stream.wait_event(produced)
-            for i, slab in enumerate(slabs):
-                lo, hi = i * S, min((i + 1) * S, nbytes)
-                if wire_fp8:
+        if self.config.verify_crc or wire_fp8:
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+        for i, slab in enumerate(slabs):
+            lo, hi = i * S, min((i + 1) * S, nbytes)
+            if wire_fp8:
+                # fp8 cast must produce the wire bytes anyway; keep fused.
+                with torch.cuda.stream(self._copy_stream):
                     crc_outs.append(
                         self._ext.pack_fp8_async(
                             t.view(-1)[lo:hi], slab[3][: hi - lo]
                         )
                     )
-                else:
-                    flat = t.view(-1).view(torch.uint8)
-                    crc_outs.append(
-                        self._ext.pack_crc_async(flat[lo:hi], slab[3][: hi - lo])
-                    )
+            else:
+                flat = t.view(-1).view(torch.uint8)
+                with torch.cuda.stream(self._copy_stream):
+                    slab[3][: hi - lo].copy_(flat[lo:hi])
+                if self.config.verify_crc:
+                    with torch.cuda.stream(self._crc_stream):
+                        crc_outs.append(self._ext.crc32_async(flat[lo:hi]))
+        with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
+        if crc_outs:
+            with torch.cuda.stream(self._crc_stream):
+                crc_done = self._crc_stream.record_event()
+            crc_done.synchronize()
         crcs = (
             [int(c[2].item()) & 0xFFFFFFFF for c in crc_outs]
             if self.config.verify_crc
@@ -261,23 +277,31 @@ class GpuDataPlane:
         wire_fp8 = man.get("wire") == "fp8e4m3"
         out = torch.empty(shape, dtype=dtype, device=self.device)
         crc_outs = []
-        with torch.cuda.stream(self._copy_stream):
-            for i, h in enumerate(handles):
-                lo, hi = i * S, min((i + 1) * S, nbytes)
-                src = self._ipc_src_view(h, hi - lo)[: hi - lo]
-                if wire_fp8:
-                    if self.config.verify_crc and crcs is not None:
-                        crc_outs.append(self._ext.crc32_async(src))
+        # Copy ∥ CRC on separate streams (see pack_to_ipc): the D2D copy runs
+        # at memory rate while the slower CRC pass verifies the same source.
+        for i, h in enumerate(handles):
+            lo, hi = i * S, min((i + 1) * S, nbytes)
+            src = self._ipc_src_view(h, hi - lo)[: hi - lo]
+            if wire_fp8:
+                with torch.cuda.stream(self._copy_stream):
                     self._ext.unpack_fp8_async(src, out.view(-1)[lo:hi])
-                else:
-                    flat = out.view(-1).view(torch.uint8)
-                    if self.config.verify_crc and crcs is not None:
-                        # Fused D2D copy + CRC in one pass over src.
-                        crc_outs.append(self._ext.pack_crc_async(src, flat[lo:hi]))
-                    else:
-                        flat[lo:hi].copy_(src)
+                if self.config.verify_crc and crcs is not None:
+                    with torch.cuda.stream(self._crc_stream):
+                        crc_outs.append(self._ext.crc32_async(src))
+            else:
+                flat = out.view(-1).view(torch.uint8)
+                with torch.cuda.stream(self._copy_stream):
+                    flat[lo:hi].copy_(src)
+                if self.config.verify_crc and crcs is not None:
+                    with torch.cuda.stream(self._crc_stream):
+                        crc_outs.append(self._ext.crc32_async(src))
+        with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
+        if crc_outs:
+            with torch.cuda.stream(self._crc_stream):
+                crc_done = self._crc_stream.record_event()
+            crc_done.synchronize()
         for i, c in enumerate(crc_outs):
             got = int(c[2].item()) & 0xFFFFFFFF
             if got != crcs[i]:
