@@ -198,7 +198,7 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
     import torch
 
     import rayfed_amd as fed
-    from rayfed_amd.parallel.fedavg import BucketedAllReducer, weighted_combine_
+    from rayfed_amd.parallel.fedavg import weighted_combine_
 
     use_gpu = _has_cuda()
     dev = f"cuda:{device}" if use_gpu else "cpu"
@@ -209,16 +209,20 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
 
     dtype = torch.bfloat16
     torch.manual_seed(0 if party == "alice" else 1)
-    # empty+uniform_ is ~5x faster than randn for bf16 on CPU and identical
-    # for bandwidth purposes (content-independent copies/collectives).
-    grads = [
-        torch.empty(s, dtype=dtype, device=dev).uniform_(-1, 1)
-        for s in llama3_8b_grad_shapes(layers, vocab)
-    ]
-    nbytes = sum(g.numel() * g.element_size() for g in grads)
-    reducer = (
-        BucketedAllReducer(group=party_group) if party_group is not None else None
-    )
+    # Flat gradient buffer with per-parameter VIEWS (the flat-grads layout
+    # production DDP keeps): the cross-party push and the intra-party
+    # all-reduce both operate on `flat` directly — no per-round repack.
+    shapes = llama3_8b_grad_shapes(layers, vocab)
+    total = sum(int(torch.prod(torch.tensor(sh))) for sh in shapes)
+    flat = torch.empty(total, dtype=dtype, device=dev).uniform_(-1, 1)
+    grads, off = [], 0
+    for sh in shapes:
+        n_el = int(torch.prod(torch.tensor(sh)))
+        grads.append(flat[off : off + n_el].view(sh))
+        off += n_el
+    nbytes = total * flat.element_size()
+    from rayfed_amd.parallel.fedavg import allreduce_flat_
+
     @fed.remote
     class Exchanger:
         """Receives the peer's averaged grads and combines with local."""
@@ -234,18 +238,11 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
         "alice": Exchanger.party("alice").remote(),
         "bob": Exchanger.party("bob").remote(),
     }
-    # One flat buffer for the cross-party push (a single big tensor rides
-    # the chunk-pipelined shm lane).
-    flat = torch.empty(nbytes // 2, dtype=dtype, device=dev)
-
     @fed.remote
     def produce(_tick):
-        if reducer is not None:
-            reducer.allreduce_(grads)  # intra-party RCCL over xGMI
-        off = 0
-        for g in grads:
-            flat[off : off + g.numel()].copy_(g.view(-1))
-            off += g.numel()
+        if party_group is not None:
+            # Intra-party RCCL over xGMI, in place on the flat buffer.
+            allreduce_flat_(flat, group=party_group)
         return flat
 
     def round_once(tick):
@@ -314,22 +311,22 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
         )
     else:
         # Member rank: mirror the leader's per-round intra-party all-reduce.
+        from rayfed_amd.parallel.fedavg import allreduce_flat_
+
         dev = f"cuda:{local_rank}" if use_gpu else "cpu"
         torch.manual_seed(rank)
-        grads = [
-            torch.empty(s, dtype=torch.bfloat16, device=dev).uniform_(-1, 1)
-            for s in llama3_8b_grad_shapes(layers, vocab)
-        ]
-        nbytes = sum(g.numel() * g.element_size() for g in grads)
-        reducer = BucketedAllReducer(group=group)
+        shapes = llama3_8b_grad_shapes(layers, vocab)
+        total = sum(int(torch.prod(torch.tensor(sh))) for sh in shapes)
+        flat = torch.empty(total, dtype=torch.bfloat16, device=dev).uniform_(-1, 1)
+        nbytes = total * 2
         for _ in range(warmup):
-            reducer.allreduce_(grads)
+            allreduce_flat_(flat, group=group)
         if use_gpu:
             torch.cuda.synchronize()
         barrier()
         t0 = time.perf_counter()
         for _ in range(steps):
-            reducer.allreduce_(grads)
+            allreduce_flat_(flat, group=group)
         if use_gpu:
             torch.cuda.synchronize()
         elapsed = time.perf_counter() - t0

@@ -105,6 +105,39 @@ class BucketedAllReducer:
             off += t.numel()
 
 
+def allreduce_flat_(
+    flat: torch.Tensor,
+    bucket_bytes: int = DEFAULT_BUCKET_BYTES,
+    group: Optional["dist.ProcessGroup"] = None,
+    average: bool = True,
+    max_inflight: int = 2,
+) -> None:
+    """In-place bucketed all-reduce of an already-flat gradient buffer
+    (the flat-grads layout real DDP stacks keep): no pack/unpack copies at
+    all — chunk views of ``flat`` reduce directly, pipelined ``max_inflight``
+    deep for xGMI ring overlap."""
+    world = dist.get_world_size(group)
+    if world == 1 or flat.numel() == 0:
+        return
+    step = max(1, bucket_bytes // flat.element_size())
+    pending = []
+    for lo in range(0, flat.numel(), step):
+        chunk = flat[lo : lo + step]
+        pending.append(
+            (dist.all_reduce(chunk, op=dist.ReduceOp.SUM, group=group,
+                             async_op=True), chunk)
+        )
+        if len(pending) > max_inflight:
+            work, done_chunk = pending.pop(0)
+            work.wait()
+            if average:
+                done_chunk.div_(world)
+    for work, chunk in pending:
+        work.wait()
+        if average:
+            chunk.div_(world)
+
+
 def allreduce_gradients(
     params_or_grads: Iterable[torch.Tensor],
     bucket_bytes: int = DEFAULT_BUCKET_BYTES,

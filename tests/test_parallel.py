@@ -140,3 +140,41 @@ def test_worker_pool_with_party_group_collective():
         assert vals == [10.0, 10.0]
     finally:
         pool.shutdown()
+
+
+def _flat_allreduce_worker(rank, world, port, q):
+    import torch
+
+    from rayfed_amd.parallel.fedavg import allreduce_flat_
+    from rayfed_amd.parallel.group import destroy_party_group, init_party_group
+
+    init_party_group(rank, world, master_port=port, backend="gloo")
+    torch.manual_seed(rank)
+    flat = torch.randn(100_000)
+    ref_inputs = []
+    for r in range(world):
+        torch.manual_seed(r)
+        ref_inputs.append(torch.randn(100_000))
+    expect = sum(ref_inputs) / world
+    allreduce_flat_(flat, bucket_bytes=64 * 1024)
+    q.put((rank, bool(torch.allclose(flat, expect, atol=1e-5))))
+    destroy_party_group()
+
+
+def test_allreduce_flat_gloo_world2():
+    """In-place chunked all-reduce of a flat grad buffer (bench fedavg path)."""
+    port = free_ports(1)[0]
+    q = _mp.Queue()
+    procs = [
+        _mp.Process(target=_flat_allreduce_worker, args=(r, 2, port, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, ok = q.get(timeout=120)
+        results[rank] = ok
+    for p in procs:
+        p.join(timeout=30)
+    assert results == {0: True, 1: True}
