@@ -62,7 +62,7 @@ class GpuDataPlane:
         self._dev_staging: List[torch.Tensor] = []
         # Device-IPC lane state: pooled hipMalloc slabs we own (handle →
         # slab) and peer handles we've opened (handle → ptr).
-        self._ipc_pool: List = []
+        self._ipc_pool: dict = {}  # size-class -> [slabs]
         self._own_ipc: dict = {}
         self._ipc_open_cache: dict = {}
 
@@ -171,22 +171,28 @@ class GpuDataPlane:
     IPC_SLAB_BYTES = 1 << 30
 
     def _ipc_get(self, nbytes: int):
-        nbytes = min(nbytes, self.IPC_SLAB_BYTES)
+        """Acquire a pooled IPC slab sized to the request's power-of-two
+        class (1 MiB .. IPC_SLAB_BYTES).  Size classes keep a payload of
+        many mid-sized tensors (a model state_dict) from pinning
+        full-size slabs per tensor — the staging footprint tracks the
+        payload, not tensor count × 1 GiB."""
+        size = 1 << max(20, min(nbytes - 1, self.IPC_SLAB_BYTES - 1).bit_length())
+        size = min(size, self.IPC_SLAB_BYTES)
         with self._lock:
-            if self._ipc_pool:
-                return self._ipc_pool.pop()
-        ptr, handle = self._ext.ipc_alloc(self.IPC_SLAB_BYTES)
-        view = self._ext.tensor_from_ptr(
-            ptr, self.IPC_SLAB_BYTES, self.device.index
-        )
-        slab = (ptr, bytes(handle), self.IPC_SLAB_BYTES, view)
+            bucket = self._ipc_pool.get(size)
+            if bucket:
+                return bucket.pop()
+        ptr, handle = self._ext.ipc_alloc(size)
+        view = self._ext.tensor_from_ptr(ptr, size, self.device.index)
+        slab = (ptr, bytes(handle), size, view)
         self._own_ipc[slab[1]] = slab
         return slab
 
     def _ipc_put(self, slab):
         with self._lock:
-            if len(self._ipc_pool) < max(self.config.staging_buffers, 20):
-                self._ipc_pool.append(slab)
+            bucket = self._ipc_pool.setdefault(slab[2], [])
+            if len(bucket) < max(self.config.staging_buffers, 20):
+                bucket.append(slab)
                 return
         self._own_ipc.pop(slab[1], None)
         self._ext.ipc_free(slab[0])
