@@ -19,6 +19,12 @@ logger = logging.getLogger(__name__)
 
 
 def _invoke_materialized(fn, args, kwargs):
+    from rayfed_amd._private import tracing
+
+    if tracing.enabled:
+        with tracing.span(getattr(fn, "__name__", "task"), "task"):
+            args, kwargs = materialize((args, kwargs))
+            return fn(*args, **kwargs)
     args, kwargs = materialize((args, kwargs))
     return fn(*args, **kwargs)
 

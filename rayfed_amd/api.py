@@ -103,6 +103,9 @@ def init(
     fed_config._clear_cached_config()
 
     setup_logger(logging_level=logging_level, party=party, job_name=job_name)
+    from rayfed_amd._private import tracing
+
+    tracing.configure(config.get("trace_file"))
     logger.info("Started rayfed-amd with %s", cluster_config)
 
     global _original_sigint_handler
@@ -207,6 +210,11 @@ def _shutdown(intended: bool = True):
 
     clear_global_context(wait_for_sending=wait_for_sending)
     barriers._cleanup_proxies()
+    from rayfed_amd._private import tracing
+
+    trace_path = tracing.flush()
+    if trace_path:
+        logger.info("Wrote trace to %s", trace_path)
     kv_mod._clear_internal_kv()
     fed_config._clear_cached_config()
 

@@ -418,6 +418,13 @@ def send(
     """Push ``data`` (an ObjectRef or a plain value) to ``dest_party`` under
     the given seq ids; ack-tracked by the cleanup manager."""
     assert _sender_service is not None, "sender proxy not started; call fed.init"
+    from rayfed_amd._private import tracing
+
+    if tracing.enabled:
+        tracing.event(
+            "send", "xsilo", dest=dest_party, up=str(upstream_seq_id),
+            down=str(downstream_seq_id), is_error=is_error,
+        )
     fut = _sender_service.send(dest_party, data, upstream_seq_id, downstream_seq_id)
     ctx = get_global_context()
     if ctx is not None:
@@ -431,6 +438,13 @@ def recv(party: str, src_party: str, upstream_seq_id, curr_seq_id) -> ObjectRef:
     """The receive barrier: an ObjectRef that resolves when the peer's push
     for (upstream_seq_id, curr_seq_id) lands in the mailbox."""
     assert _receiver_service is not None, "receiver proxy not started; call fed.init"
+    from rayfed_amd._private import tracing
+
+    if tracing.enabled:
+        tracing.event(
+            "recv", "xsilo", src=src_party, up=str(upstream_seq_id),
+            down=str(curr_seq_id),
+        )
     fut = _receiver_service.get_data(src_party, upstream_seq_id, curr_seq_id)
     return ObjectRef(fut)
 

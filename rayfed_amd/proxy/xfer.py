@@ -131,6 +131,15 @@ class XferSenderService:
         )
 
     def _send_blocking(self, dest_party, data, up, down) -> bool:
+        from rayfed_amd._private import tracing
+
+        if tracing.enabled:
+            with tracing.span("xfer.send", "xsilo", dest=dest_party, up=up,
+                              down=down):
+                return self._send_blocking_inner(dest_party, data, up, down)
+        return self._send_blocking_inner(dest_party, data, up, down)
+
+    def _send_blocking_inner(self, dest_party, data, up, down) -> bool:
         t0 = time.perf_counter()
         err = True
         nbytes = 0

@@ -1,0 +1,56 @@
+"""Chrome-trace export (new observability subsystem; the reference has
+none — SURVEY.md §5)."""
+import json
+import os
+
+import rayfed_amd as fed
+from tests._util import run_parties
+
+
+def _driver_traced(party, addresses, trace_dir):
+    fed.init(
+        addresses=addresses,
+        party=party,
+        config={"trace_file": os.path.join(trace_dir, "trace.json")},
+        logging_level="warning",
+    )
+
+    @fed.remote
+    def make(v):
+        return v * 2
+
+    @fed.remote
+    def add(a, b):
+        return a + b
+
+    x = make.party("alice").remote(3)
+    y = make.party("bob").remote(4)
+    s = add.party("bob").remote(x, y)
+    assert fed.get(s) == 14
+    fed.shutdown()
+
+
+def test_trace_file_written_and_parsable(tmp_path):
+    run_parties(_driver_traced, args=(str(tmp_path),))
+    files = [p for p in os.listdir(tmp_path) if p.startswith("trace.")]
+    assert len(files) == 2, files  # one per party process
+    cats = set()
+    names = set()
+    for fname in files:
+        with open(tmp_path / fname) as f:
+            data = json.load(f)
+        events = data["traceEvents"]
+        assert events, fname
+        for e in events:
+            cats.add(e["cat"])
+            names.add(e["name"])
+            assert {"name", "cat", "ph", "ts", "pid", "tid"} <= set(e)
+    assert "xsilo" in cats and "task" in cats
+    assert {"send", "recv", "xfer.send"} <= names
+
+
+def test_tracing_disabled_by_default(tmp_path):
+    from rayfed_amd._private import tracing
+
+    assert tracing.enabled is False
+    tracing.event("noop", "x")  # must be a no-op, not an error
