@@ -87,7 +87,12 @@ class XferSenderService:
         self._party = party
         self._job_name = job_name
         self._proxy_config = proxy_config
-        self._client = _load_xfer().XferClient(job_name)
+        xfer = _load_xfer()
+        # Two connections per destination: control frames must not queue
+        # behind a multi-GiB defer-ack consume on the bulk lane (a float
+        # broadcast measured 100 ms stuck behind a 16 GB frame's consume).
+        self._client_ctl = xfer.XferClient(job_name)
+        self._client_bulk = xfer.XferClient(job_name)
         self._pool = ThreadPoolExecutor(max_workers=8, thread_name_prefix="xfer-send")
         self._retry = _Retry(
             getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
@@ -182,8 +187,10 @@ class XferSenderService:
             try:
                 nbytes = sum(len(p) for p in body_parts)
                 host, port = self._addresses[dest_party].rsplit(":", 1)
+                bulk = defer_ack or nbytes > (1 << 20)
                 code, result = self._send_with_retry(
-                    host, int(port), up, down, body_parts, defer_ack
+                    self._client_bulk if bulk else self._client_ctl,
+                    host, int(port), up, down, body_parts, defer_ack,
                 )
             finally:
                 tensor_codec.release_parts(extras)
@@ -203,7 +210,7 @@ class XferSenderService:
                 edge = self._edges.setdefault(dest_party, self._edge_cls())
                 edge.record(nbytes, secs, err)
 
-    def _send_with_retry(self, host, port, up, down, parts, defer_ack):
+    def _send_with_retry(self, client, host, port, up, down, parts, defer_ack):
         deadline = time.monotonic() + self._timeout_s
         backoff = self._retry.initial_backoff
         attempt = 0
@@ -211,7 +218,7 @@ class XferSenderService:
             attempt += 1
             try:
                 remaining = max(0.001, deadline - time.monotonic())
-                return self._client.send(
+                return client.send(
                     host, port, up, down, parts, defer_ack, remaining
                 )
             except RuntimeError as e:
@@ -234,7 +241,8 @@ class XferSenderService:
 
     def stop(self):
         self._pool.shutdown(wait=False, cancel_futures=True)
-        self._client.close_all()
+        self._client_ctl.close_all()
+        self._client_bulk.close_all()
 
 
 class XferReceiverService:
