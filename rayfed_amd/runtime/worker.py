@@ -185,19 +185,30 @@ class DeviceWorkerPool:
                 break
             task_id, is_error, value = cloudpickle.loads(msg)
             with self._fut_lock:
-                fut = self._futures.pop(task_id, None)
-            if fut is None:
+                entry = self._futures.pop(task_id, None)
+            if entry is None:
                 continue
+            fut, _w = entry
             if is_error:
                 fut.set_exception(value)
             else:
                 fut.set_result(value)
+        # Worker gone (crash or shutdown): fail its outstanding futures so
+        # callers never hang on a dead process.
+        with self._fut_lock:
+            dead = [tid for tid, (_f, w) in self._futures.items() if w == wid]
+            entries = [self._futures.pop(tid) for tid in dead]
+        for fut, _w in entries:
+            if not fut.done():
+                fut.set_exception(
+                    RuntimeError(f"device worker {wid} exited unexpectedly")
+                )
 
     def _send(self, wid: int, op: str, payload) -> ObjectRef:
         task_id = next(self._task_counter)
         fut: Future = Future()
         with self._fut_lock:
-            self._futures[task_id] = fut
+            self._futures[task_id] = (fut, wid)
         msg = cloudpickle.dumps((op, task_id, payload))
         with self._send_locks[wid]:
             self._conns[wid].send_bytes(msg)

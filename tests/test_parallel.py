@@ -178,3 +178,22 @@ def test_allreduce_flat_gloo_world2():
     for p in procs:
         p.join(timeout=30)
     assert results == {0: True, 1: True}
+
+
+def _die():
+    import os
+
+    os._exit(17)
+
+
+def test_worker_death_fails_outstanding_futures():
+    """A crashed worker must fail its pending refs, not hang callers."""
+    from rayfed_amd.runtime.worker import DeviceWorkerPool
+
+    pool = DeviceWorkerPool(devices=[None])
+    try:
+        ref = pool.submit(0, _die)
+        with pytest.raises(RuntimeError, match="exited unexpectedly"):
+            ref.result(60)
+    finally:
+        pool.shutdown()
