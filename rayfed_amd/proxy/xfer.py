@@ -332,7 +332,18 @@ class XferReceiverService:
             return fut
 
         def _wait():
-            b = self._server.get_data(up, down, 600.0)
+            # Parity with the reference's recv semantics: wait indefinitely
+            # for the peer's push (failure paths deliver an error object on
+            # the same seq ids instead of leaving this hanging).  Slice the
+            # C++ wait so server stop still unblocks promptly.
+            while True:
+                try:
+                    b = self._server.get_data(up, down, 60.0)
+                    break
+                except RuntimeError as e:
+                    if "timeout" in str(e):
+                        continue
+                    raise  # server stopped
             return self._take(up, down, b)
 
         return self._pool.submit(_wait)
