@@ -23,10 +23,15 @@ def test_plain_object_no_tensors():
 
 
 @pytest.mark.parametrize(
-    "dtype", [torch.float32, torch.bfloat16, torch.float16, torch.int64, torch.uint8]
+    "dtype",
+    [torch.float32, torch.float64, torch.bfloat16, torch.float16,
+     torch.int64, torch.uint8, torch.bool, torch.complex64],
 )
 def test_single_tensor_roundtrip(dtype):
-    t = (torch.arange(257) % 128).to(dtype)
+    if dtype is torch.bool:
+        t = (torch.arange(257) % 2).to(dtype)
+    else:
+        t = (torch.arange(257) % 128).to(dtype)
     extras, out = _roundtrip(t)
     assert len(extras["tensors"]) == 1
     man = extras["tensors"][0]
