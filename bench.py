@@ -18,8 +18,11 @@ for N>1 via ``python -m torch.distributed.run --nproc-per-node N`` — one
 rank per GPU, rank r < N/2 drives alice of lane r, rank r >= N/2 drives bob
 of lane r-N/2.  Rank 0 prints ONE JSON line; elapsed is the MAX over ranks.
 
-Modes: ``tiny`` (headline), ``push`` (BASELINE config 3: 4 GiB bf16 tensor
-push alice→bob, GB/s — reported in the JSON config block when run).
+Modes: ``tiny`` (headline; ``--parties 3 --tls`` for BASELINE config 5),
+``push`` (config 3: 4 GiB bf16 tensor push alice→bob over the device-IPC
+lane; ``RAYFED_BENCH_WIRE_FP8=1`` for fp8 wire), ``fedavg`` (config 4:
+Llama-3-8B gradients, intra-party RCCL + cross-party exchange + HIP
+combine).
 """
 from __future__ import annotations
 
