@@ -7,7 +7,7 @@ snapshots to GPU boxes (it must live in-tree, not in site-packages).
 """
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -16,7 +16,7 @@ from torch.utils import cpp_extension  # noqa: E402
 setup(
     name="rayfed_amd",
     version="0.1.0",
-    packages=["rayfed_amd"],
+    packages=find_packages(include=["rayfed_amd", "rayfed_amd.*"]),
     ext_modules=[
         cpp_extension.CUDAExtension(
             name="rayfed_amd._hip",
