@@ -263,6 +263,99 @@ class GpuDataPlane:
 
         return [s[1] for s in slabs], man, crcs, release
 
+    def pack_group_to_ipc(self, tensors: List["torch.Tensor"]):
+        """Arena-pack several mid-size tensors into SHARED IPC slabs
+        (state_dict-style payloads: 291 Llama tensors → ~16 slabs instead of
+        291 handles).  Tensors never straddle a slab boundary.  Returns
+        (group_manifest, per_tensor_fields, release_fn) where
+        group_manifest = {"slabs": [handles], "slab_bytes": S} and each
+        per-tensor entry is {"slab": idx, "off": o, ["crc32": c]}."""
+        self._bind_device()
+        S = self.IPC_SLAB_BYTES
+        flats = []
+        for t in tensors:
+            t = t.detach()
+            if not t.is_contiguous():
+                t = t.contiguous()
+            flats.append(t.view(-1).view(torch.uint8))
+        # Place tensors: bump offset, advance slab when one would straddle.
+        placements = []
+        slab_sizes: List[int] = []
+        cur = 0
+        for f in flats:
+            n = f.numel()
+            if not slab_sizes or cur + n > S:
+                slab_sizes.append(0)
+                cur = 0
+            placements.append((len(slab_sizes) - 1, cur))
+            cur += n
+            slab_sizes[-1] = cur
+        slabs = [self._ipc_get(sz) for sz in slab_sizes]
+        produced = torch.cuda.current_stream(self.device).record_event()
+        with torch.cuda.stream(self._copy_stream):
+            self._copy_stream.wait_event(produced)
+        crc_outs = []
+        if self.config.verify_crc:
+            with torch.cuda.stream(self._crc_stream):
+                self._crc_stream.wait_event(produced)
+        for f, (si, off) in zip(flats, placements):
+            n = f.numel()
+            with torch.cuda.stream(self._copy_stream):
+                slabs[si][3][off : off + n].copy_(f)
+            if self.config.verify_crc:
+                with torch.cuda.stream(self._crc_stream):
+                    crc_outs.append(self._ext.crc32_async(f))
+        with torch.cuda.stream(self._copy_stream):
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        if crc_outs:
+            with torch.cuda.stream(self._crc_stream):
+                crc_done = self._crc_stream.record_event()
+            crc_done.synchronize()
+        group = {"slabs": [s[1] for s in slabs], "slab_bytes": S}
+        fields = []
+        for i, (si, off) in enumerate(placements):
+            f = {"slab": si, "off": off}
+            if self.config.verify_crc:
+                f["crc32"] = int(crc_outs[i][2].item()) & 0xFFFFFFFF
+            fields.append(f)
+
+        def release():
+            for s in slabs:
+                self._ipc_put(s)
+
+        return group, fields, release
+
+    def unpack_from_ipc_group(self, group, man, dtype, shape):
+        """Receiver of an arena-packed tensor: D2D from (slab, off) with the
+        CRC verify overlapped on the crc stream."""
+        self._bind_device()
+        nbytes = man["nbytes"]
+        handle = bytes(group["slabs"][man["slab"]])
+        off = man["off"]
+        src = self._ipc_src_view(handle, off + nbytes)[off : off + nbytes]
+        out = torch.empty(shape, dtype=dtype, device=self.device)
+        crc_expect = man.get("crc32")
+        crc_out = None
+        flat = out.view(-1).view(torch.uint8)
+        with torch.cuda.stream(self._copy_stream):
+            flat.copy_(src)
+            done = self._copy_stream.record_event()
+        if self.config.verify_crc and crc_expect is not None:
+            with torch.cuda.stream(self._crc_stream):
+                crc_out = self._ext.crc32_async(src)
+                crc_done = self._crc_stream.record_event()
+        done.synchronize()
+        if crc_out is not None:
+            crc_done.synchronize()
+            got = int(crc_out[2].item()) & 0xFFFFFFFF
+            if got != crc_expect:
+                raise ValueError(
+                    f"GPU tensor CRC mismatch (ipc group): expected "
+                    f"{crc_expect:#x}, got {got:#x}"
+                )
+        return out
+
     def _ipc_src_view(self, handle: bytes, nbytes: int):
         own = self._own_ipc.get(handle)
         if own is not None:  # same-process loopback: use the local mapping

@@ -125,12 +125,62 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
     manifests: List[Dict] = []
     parts: List[memoryview] = [memoryview(skeleton)]
     releases: List = []
+
+    # Route every tensor first so mid-size device tensors can be ARENA-packed
+    # together into shared IPC slabs (a state_dict push ships ~16 slab
+    # handles instead of one per tensor).
+    import os as _os
+
+    ipc_on = _os.environ.get("RAYFED_IPC", "1") != "0"
+    if shm:
+        from rayfed_amd.ops import shm_pool
+
+        shm_min = shm_pool.SHM_MIN_BYTES
+    routes = []
     for t in pickler.tensors:
         if torch is None:
             raise RuntimeError("torch payload without torch installed")
         dtype = _DTYPE_TO_STR.get(t.dtype)
         if dtype is None:
             raise TypeError(f"unsupported tensor dtype {t.dtype}")
+        nbytes = t.numel() * t.element_size()
+        route = "payload"
+        if shm and nbytes >= shm_min:
+            if gpu_plane is not None and t.device.type == "cuda":
+                wire_fp8 = (
+                    gpu_plane.config.wire_dtype == "fp8e4m3"
+                    and t.dtype == torch.bfloat16
+                )
+                if ipc_on and not wire_fp8 and nbytes <= gpu_plane.IPC_SLAB_BYTES:
+                    route = "ipc_group"
+                elif ipc_on:
+                    route = "ipc"
+                elif not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
+                    route = "shm_chunked"
+                else:
+                    route = "shm_gpu"
+            else:
+                route = "shm_cpu"
+        routes.append(route)
+
+    group_members = [i for i, r in enumerate(routes) if r == "ipc_group"]
+    if len(group_members) == 1:
+        routes[group_members[0]] = "ipc"  # single tensor: no group overhead
+        group_members = []
+    group_fields = {}
+    if group_members:
+        group_man, fields, release = gpu_plane.pack_group_to_ipc(
+            [pickler.tensors[i] for i in group_members]
+        )
+        releases.append(release)
+        for i, f in zip(group_members, fields):
+            group_fields[i] = f
+        group_extras = group_man
+    else:
+        group_extras = None
+
+    for idx, t in enumerate(pickler.tensors):
+        dtype = _DTYPE_TO_STR[t.dtype]
         man = {
             "dtype": dtype,
             "shape": list(t.shape),
@@ -138,64 +188,68 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
             "nbytes": t.numel() * t.element_size(),
         }
         nbytes = man["nbytes"]
-        use_shm = False
-        if shm:
-            from rayfed_amd.ops import shm_pool
-
-            use_shm = nbytes >= shm_pool.SHM_MIN_BYTES
-        if use_shm:
-            # Same-host lane: bytes land in a pooled /dev/shm segment; only
-            # the (segment, offset) reference rides the socket.  The segment
-            # is recycled when the receiver's ack releases it.
-            from rayfed_amd.ops import shm_pool
-
-            if gpu_plane is not None and t.device.type == "cuda":
-                import os as _os
-
-                wire_fp8 = (
-                    gpu_plane.config.wire_dtype == "fp8e4m3"
-                    and t.dtype == torch.bfloat16
-                )
-                if _os.environ.get("RAYFED_IPC", "1") != "0":
-                    # Device-IPC lane: no host bounce at all (same node).
-                    _, man_fields, _crcs, release = gpu_plane.pack_to_ipc(t)
-                    man.update(man_fields)  # ipc_slabs/slab_bytes/ipc_crcs/wire
-                    if wire_fp8:
-                        man["nbytes"] = t.numel()
-                    releases.append(release)
-                    manifests.append(man)
-                    continue
-                if not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
-                    # Chunk pipeline: D2H publishes progress as it goes; the
-                    # frame leaves before the DMA finishes and the receiver's
-                    # H2D overlaps it.
-                    tc = t.detach()
-                    if not tc.is_contiguous():
-                        tc = tc.contiguous()
-                    seg, man_fields, release = gpu_plane.pack_to_shm_chunked(tc)
-                    man.update(man_fields)
-                else:
-                    if wire_fp8:
-                        man["wire"] = "fp8e4m3"
-                        man["nbytes"] = t.numel()
-                    seg, crc, release = gpu_plane.pack_to_shm(t)
-                    man["crc32"] = crc
-            else:
-                pool = shm_pool.get_send_pool()
-                seg = pool.acquire(nbytes)
-                raw_cpu = _tensor_bytes_cpu(t)
-                seg.array[: len(raw_cpu)] = memoryview(raw_cpu)
-                man["nbytes"] = len(raw_cpu)
-                if gpu_plane is not None and gpu_plane.config.verify_crc:
-                    import zlib
-
-                    man["crc32"] = zlib.crc32(raw_cpu) & 0xFFFFFFFF
-                release = lambda s=seg, p=pool: p.release(s)  # noqa: E731
+        route = routes[idx]
+        if route == "ipc_group":
+            man["ipcg"] = True
+            man.update(group_fields[idx])
+            manifests.append(man)
+            continue
+        if route == "ipc":
+            wire_fp8 = (
+                gpu_plane.config.wire_dtype == "fp8e4m3"
+                and t.dtype == torch.bfloat16
+            )
+            _, man_fields, _crcs, release = gpu_plane.pack_to_ipc(t)
+            man.update(man_fields)  # ipc_slabs/slab_bytes/ipc_crcs/wire
+            if wire_fp8:
+                man["nbytes"] = t.numel()
+            releases.append(release)
+            manifests.append(man)
+            continue
+        if route == "shm_chunked":
+            tc = t.detach()
+            if not tc.is_contiguous():
+                tc = tc.contiguous()
+            seg, man_fields, release = gpu_plane.pack_to_shm_chunked(tc)
+            man.update(man_fields)
             man["shm"] = seg.name
             man["shm_off"] = 0
             releases.append(release)
             manifests.append(man)
-            continue  # no payload part for shm-borne tensors
+            continue
+        if route == "shm_gpu":
+            wire_fp8 = (
+                gpu_plane.config.wire_dtype == "fp8e4m3"
+                and t.dtype == torch.bfloat16
+            )
+            if wire_fp8:
+                man["wire"] = "fp8e4m3"
+                man["nbytes"] = t.numel()
+            seg, crc, release = gpu_plane.pack_to_shm(t)
+            man["crc32"] = crc
+            man["shm"] = seg.name
+            man["shm_off"] = 0
+            releases.append(release)
+            manifests.append(man)
+            continue
+        if route == "shm_cpu":
+            from rayfed_amd.ops import shm_pool
+
+            pool = shm_pool.get_send_pool()
+            seg = pool.acquire(nbytes)
+            raw_cpu = _tensor_bytes_cpu(t)
+            seg.array[: len(raw_cpu)] = memoryview(raw_cpu)
+            man["nbytes"] = len(raw_cpu)
+            if gpu_plane is not None and gpu_plane.config.verify_crc:
+                import zlib
+
+                man["crc32"] = zlib.crc32(raw_cpu) & 0xFFFFFFFF
+            releases.append(lambda s=seg, p=pool: p.release(s))
+            man["shm"] = seg.name
+            man["shm_off"] = 0
+            manifests.append(man)
+            continue
+        # payload route
         if gpu_plane is not None and t.device.type == "cuda":
             if (
                 gpu_plane.config.wire_dtype == "fp8e4m3"
@@ -216,6 +270,8 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
         manifests.append(man)
         parts.append(memoryview(raw))
     extras = {"skel": len(skeleton), "tensors": manifests}
+    if group_extras is not None:
+        extras["ipc_group"] = group_extras
     if releases:
         extras["_releases"] = releases  # stripped before hitting the wire
     return extras, parts
@@ -294,6 +350,19 @@ def decode(
     tensors: List[Any] = []
     for man in extras["tensors"]:
         nbytes = man["nbytes"]
+        if man.get("ipcg"):
+            if gpu_plane is None:
+                raise RuntimeError(
+                    "received a device-IPC tensor but no GPU data plane is "
+                    "attached (set RAYFED_IPC=0 on the sender for CPU peers)"
+                )
+            tensors.append(
+                gpu_plane.unpack_from_ipc_group(
+                    extras["ipc_group"], man, _STR_TO_DTYPE[man["dtype"]],
+                    man["shape"],
+                )
+            )
+            continue
         if "ipc_slabs" in man:
             if gpu_plane is None:
                 raise RuntimeError(

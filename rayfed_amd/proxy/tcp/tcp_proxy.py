@@ -336,7 +336,7 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
                     # request body until the reader consumes it.
                     kind, header, payload = frames.decode_frame(body[8:])
                     if kind == frames.KIND_TENSOR and any(
-                        "shm" in m or "ipc_slabs" in m
+                        "shm" in m or "ipc_slabs" in m or m.get("ipcg")
                         for m in header.get("tensors", ())
                     ):
                         # shm lane: consume (H2D + CRC) BEFORE acking — the
@@ -376,7 +376,8 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
             obj = await loop.run_in_executor(
                 None,
                 tensor_codec.decode,
-                {k: header[k] for k in ("skel", "tensors")},
+                {k: header[k] for k in ("skel", "tensors", "ipc_group")
+                 if k in header},
                 memoryview(payload),
                 self.gpu_plane,
                 self._mailbox._allowed_list,

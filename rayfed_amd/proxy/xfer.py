@@ -174,7 +174,8 @@ class XferSenderService:
                     }
                     header.update(wire_header)
                     defer_ack = any(
-                        "shm" in m or "ipc_slabs" in m for m in extras["tensors"]
+                        "shm" in m or "ipc_slabs" in m or m.get("ipcg")
+                        for m in extras["tensors"]
                     )
                     body_parts = [
                         frames.encode_frame_prefix(frames.KIND_TENSOR, header)
@@ -289,7 +290,8 @@ class XferReceiverService:
         try:
             kind, header, payload = frames.decode_frame(body)
             obj = tensor_codec.decode(
-                {k: header[k] for k in ("skel", "tensors")},
+                {k: header[k] for k in ("skel", "tensors", "ipc_group")
+                 if k in header},
                 memoryview(payload),
                 self.gpu_plane,
                 self._allowed_list,

@@ -382,3 +382,32 @@ def test_ipc_size_class_pooling(plane):
     handles1 = {bytes(m["ipc_slabs"][0]) for m in extras["tensors"]}
     tensor_codec.release_parts(extras2)
     assert handles2 == handles1
+
+
+@needs_gpu
+def test_ipc_arena_group_roundtrip(plane):
+    """state_dict-style payload: mid-size tensors arena-pack into shared
+    slabs (ipcg manifests + one ipc_group); decode restores exactly."""
+    from rayfed_amd.ops import tensor_codec
+
+    state = {}
+    torch.manual_seed(5)
+    for i in range(12):
+        n = (1 << 20) + i * 4096  # ~1 MiB each, distinct sizes
+        state[f"layer{i}.weight"] = torch.randn(n // 4, device="cuda")
+    state["small.bias"] = torch.randn(64, device="cuda")  # payload route
+    extras, parts = tensor_codec.encode(state, plane, shm=True)
+    grouped = [m for m in extras["tensors"] if m.get("ipcg")]
+    assert len(grouped) == 12
+    assert "ipc_group" in extras
+    assert len(extras["ipc_group"]["slabs"]) <= 2  # shared slabs, not 12
+    assert all("crc32" in m for m in grouped)
+    payload = b"".join(bytes(p) for p in parts)
+    out = tensor_codec.decode(extras, memoryview(payload), plane, None)
+    tensor_codec.release_parts(extras)
+    for k, v in state.items():
+        assert torch.equal(out[k], v), k
+    # Group slabs recycle on re-encode.
+    extras2, _ = tensor_codec.encode(state, plane, shm=True)
+    tensor_codec.release_parts(extras2)
+    assert extras2["ipc_group"]["slabs"] == extras["ipc_group"]["slabs"]
