@@ -11,10 +11,13 @@ overhead).  Run one process per party:
 ``bench.py`` at the repo root wraps the same loop under the driver's
 measurement contract; this script is the interactive/parity harness.
 """
+import os
 import sys
 import time
 
-import rayfed_amd as fed
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import rayfed_amd as fed  # noqa: E402
 
 
 @fed.remote
