@@ -15,9 +15,14 @@
 // flags bit 0 (DEFER_ACK): consume via the Python callback BEFORE acking
 // (the shm lane's ack licenses segment recycling).
 //
-// Scope: plaintext only — TLS jobs use the Python asyncio transport
-// (selected automatically in rayfed_amd.proxy.barriers).  Thread-per-
-// connection: a federation has a handful of parties, not thousands.
+// TLS (OpenSSL, mutual auth optional) is supported on both roles.  The
+// plaintext client pipelines requests on shared sockets (reader thread per
+// connection); the TLS client instead uses a small pool of exclusive
+// connections (one write+ack exchange at a time per connection) because an
+// OpenSSL SSL* is not safe for a concurrent reader+writer pair.
+// Thread-per-connection server: a federation has a handful of parties.
+#include <openssl/err.h>
+#include <openssl/ssl.h>
 #include <pybind11/functional.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -104,18 +109,111 @@ static void set_sock_opts(int fd) {
   setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
 
-struct Ack {
-  uint16_t code;
-  std::string result;
+static std::string ssl_err() {
+  char buf[256];
+  ERR_error_string_n(ERR_get_error(), buf, sizeof(buf));
+  return std::string(buf);
+}
+
+// Byte stream over a plain fd or an SSL session.
+struct Stream {
+  int fd = -1;
+  SSL* ssl = nullptr;
+
+  bool read_all(char* data, size_t n) {
+    if (!ssl) return ::read_all(fd, data, n);
+    while (n) {
+      int r = SSL_read(ssl, data, (int)std::min(n, (size_t)1 << 20));
+      if (r <= 0) return false;
+      data += r;
+      n -= (size_t)r;
+    }
+    return true;
+  }
+
+  void write_all(const char* data, size_t n) {
+    if (!ssl) return ::write_all(fd, data, n);
+    while (n) {
+      int w = SSL_write(ssl, data, (int)std::min(n, (size_t)1 << 20));
+      if (w <= 0) throw std::runtime_error("SSL_write: " + ssl_err());
+      data += w;
+      n -= (size_t)w;
+    }
+  }
+
+  void write_iov(std::vector<iovec> iov) {
+    if (!ssl) return ::writev_all(fd, std::move(iov));
+    for (auto& v : iov) write_all((const char*)v.iov_base, v.iov_len);
+  }
+
+  void close_free() {
+    if (ssl) {
+      SSL_free(ssl);
+      ssl = nullptr;
+    }
+    if (fd >= 0) {
+      ::close(fd);
+      fd = -1;
+    }
+  }
 };
+
+static SSL_CTX* make_server_ctx(const std::string& cert, const std::string& key,
+                                const std::string& ca) {
+  SSL_CTX* ctx = SSL_CTX_new(TLS_server_method());
+  if (!ctx) throw std::runtime_error("SSL_CTX_new failed");
+  SSL_CTX_set_min_proto_version(ctx, TLS1_2_VERSION);
+  if (SSL_CTX_use_certificate_chain_file(ctx, cert.c_str()) != 1 ||
+      SSL_CTX_use_PrivateKey_file(ctx, key.c_str(), SSL_FILETYPE_PEM) != 1) {
+    SSL_CTX_free(ctx);
+    throw std::runtime_error("server cert/key load failed: " + ssl_err());
+  }
+  if (!ca.empty()) {
+    if (SSL_CTX_load_verify_locations(ctx, ca.c_str(), nullptr) != 1) {
+      SSL_CTX_free(ctx);
+      throw std::runtime_error("server CA load failed: " + ssl_err());
+    }
+    SSL_CTX_set_verify(
+        ctx, SSL_VERIFY_PEER | SSL_VERIFY_FAIL_IF_NO_PEER_CERT, nullptr);
+  }
+  return ctx;
+}
+
+static SSL_CTX* make_client_ctx(const std::string& ca, const std::string& cert,
+                                const std::string& key) {
+  SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
+  if (!ctx) throw std::runtime_error("SSL_CTX_new failed");
+  SSL_CTX_set_min_proto_version(ctx, TLS1_2_VERSION);
+  if (!ca.empty() && SSL_CTX_load_verify_locations(ctx, ca.c_str(), nullptr) != 1) {
+    SSL_CTX_free(ctx);
+    throw std::runtime_error("client CA load failed: " + ssl_err());
+  }
+  SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER, nullptr);
+  if (!cert.empty() && !key.empty()) {
+    if (SSL_CTX_use_certificate_chain_file(ctx, cert.c_str()) != 1 ||
+        SSL_CTX_use_PrivateKey_file(ctx, key.c_str(), SSL_FILETYPE_PEM) != 1) {
+      SSL_CTX_free(ctx);
+      throw std::runtime_error("client cert/key load failed: " + ssl_err());
+    }
+  }
+  return ctx;
+}
 
 // ------------------------------------------------------------------- server
 class XferServer {
  public:
-  XferServer(int port, std::string job_name)
-      : job_(std::move(job_name)), port_(port) {}
+  XferServer(int port, std::string job_name, std::string tls_cert = "",
+             std::string tls_key = "", std::string tls_ca = "")
+      : job_(std::move(job_name)), port_(port) {
+    if (!tls_cert.empty()) {
+      ssl_ctx_ = make_server_ctx(tls_cert, tls_key, tls_ca);
+    }
+  }
 
-  ~XferServer() { stop(); }
+  ~XferServer() {
+    stop();
+    if (ssl_ctx_) SSL_CTX_free(ssl_ctx_);
+  }
 
   // consume_cb(body_bytes) -> int code; called (with the GIL) for
   // DEFER_ACK frames only.  Delivery happens before the ack is written.
@@ -232,20 +330,31 @@ class XferServer {
       std::lock_guard<std::mutex> lk(conn_mu_);
       conn_fds_.push_back(fd);
       conn_threads_.emplace_back([this, fd] { conn_loop(fd); });
+
     }
   }
 
   void conn_loop(int fd) {
+    Stream st;
+    st.fd = fd;
+    if (ssl_ctx_) {
+      st.ssl = SSL_new(ssl_ctx_);
+      SSL_set_fd(st.ssl, fd);
+      if (SSL_accept(st.ssl) != 1) {
+        st.close_free();
+        return;
+      }
+    }
     std::vector<char> buf;
     while (running_) {
       char head[16];
-      if (!read_all(fd, head, 16)) break;
+      if (!st.read_all(head, 16)) break;
       uint64_t total, req_id;
       memcpy(&total, head, 8);
       memcpy(&req_id, head + 8, 8);
       if (total < 12 || total > (64ull << 30)) break;  // sane bounds
       buf.resize(total - 8);  // everything after req_id
-      if (!read_all(fd, buf.data(), buf.size())) break;
+      if (!st.read_all(buf.data(), buf.size())) break;
       uint8_t flags = (uint8_t)buf[0];
       uint8_t job_len = (uint8_t)buf[1];
       uint8_t up_len = (uint8_t)buf[2];
@@ -288,13 +397,13 @@ class XferServer {
       memcpy(ack + 4, &req_id, 8);
       memcpy(ack + 12, &code, 2);
       try {
-        write_all(fd, ack, 14);
-        if (!result.empty()) write_all(fd, result.data(), result.size());
+        st.write_all(ack, 14);
+        if (!result.empty()) st.write_all(result.data(), result.size());
       } catch (...) {
         break;
       }
     }
-    ::close(fd);
+    st.close_free();
   }
 
   std::string job_;
@@ -311,13 +420,24 @@ class XferServer {
   std::map<std::string, std::string> mail_;
   std::atomic<uint64_t> recv_count_{0};
   py::object consume_cb_;
+  SSL_CTX* ssl_ctx_ = nullptr;
 };
 
 // ------------------------------------------------------------------- client
 class XferClient {
  public:
-  XferClient(std::string job_name) : job_(std::move(job_name)) {}
-  ~XferClient() { close_all(); }
+  XferClient(std::string job_name, std::string tls_ca = "",
+             std::string tls_cert = "", std::string tls_key = "",
+             std::string server_name = "")
+      : job_(std::move(job_name)), server_name_(std::move(server_name)) {
+    if (!tls_ca.empty() || !tls_cert.empty()) {
+      ssl_ctx_ = make_client_ctx(tls_ca, tls_cert, tls_key);
+    }
+  }
+  ~XferClient() {
+    close_all();
+    if (ssl_ctx_) SSL_CTX_free(ssl_ctx_);
+  }
 
   // Blocking send with ack round trip; GIL released around I/O.
   // parts: list of buffer-likes written scatter-gather (no join copy).
@@ -352,6 +472,11 @@ class XferClient {
 
     int code;
     std::string result;
+    if (ssl_ctx_) {
+      py::gil_scoped_release release;
+      return send_tls(host, port, preamble, views, total, timeout_s,
+                      result_out);
+    }
     {
       py::gil_scoped_release release;
       std::shared_ptr<Conn> conn_sp = get_conn(host, port);
@@ -396,6 +521,102 @@ class XferClient {
     return code;
   }
 
+  int send_tls(const std::string& host, int port, const std::string& preamble,
+               const std::vector<std::pair<const char*, size_t>>& views,
+               uint64_t total, double timeout_s, std::string* result_out) {
+    std::string key = host + ":" + std::to_string(port);
+    std::unique_ptr<Stream> st = acquire_tls_conn(key, host, port);
+    uint64_t req_id = tls_req_id_.fetch_add(1);
+    try {
+      char head[16];
+      memcpy(head, &total, 8);
+      memcpy(head + 8, &req_id, 8);
+      std::vector<iovec> iov;
+      iov.push_back({head, 16});
+      iov.push_back({(void*)preamble.data(), preamble.size()});
+      for (auto& v : views) iov.push_back({(void*)v.first, v.second});
+      st->write_iov(std::move(iov));
+      char ahead[14];
+      if (!st->read_all(ahead, 14))
+        throw std::runtime_error("connection broken awaiting ack");
+      uint32_t len;
+      uint64_t rid;
+      uint16_t code;
+      memcpy(&len, ahead, 4);
+      memcpy(&rid, ahead + 4, 8);
+      memcpy(&code, ahead + 12, 2);
+      std::string result;
+      if (len > 10) {
+        result.resize(len - 10);
+        if (!st->read_all(result.data(), result.size()))
+          throw std::runtime_error("connection broken reading ack body");
+      }
+      if (rid != req_id) throw std::runtime_error("ack id mismatch on TLS lane");
+      release_tls_conn(key, std::move(st));
+      if (result_out) *result_out = result;
+      return code;
+    } catch (...) {
+      st->close_free();
+      throw;
+    }
+  }
+
+  std::unique_ptr<Stream> acquire_tls_conn(const std::string& key,
+                                           const std::string& host, int port) {
+    {
+      std::lock_guard<std::mutex> lk(tls_mu_);
+      auto& pool = tls_pool_[key];
+      if (!pool.empty()) {
+        auto st = std::move(pool.back());
+        pool.pop_back();
+        return st;
+      }
+    }
+    int fd = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (fd < 0) throw std::runtime_error("socket() failed");
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_port = htons((uint16_t)port);
+    if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1) {
+      hostent* he = gethostbyname(host.c_str());
+      if (!he) {
+        ::close(fd);
+        throw std::runtime_error("resolve failed: " + host);
+      }
+      memcpy(&addr.sin_addr, he->h_addr, sizeof(addr.sin_addr));
+    }
+    if (connect(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
+      ::close(fd);
+      throw std::runtime_error(std::string("connect failed: ") +
+                               strerror(errno));
+    }
+    set_sock_opts(fd);
+    auto st = std::make_unique<Stream>();
+    st->fd = fd;
+    st->ssl = SSL_new(ssl_ctx_);
+    SSL_set_fd(st->ssl, fd);
+    if (!server_name_.empty()) {
+      SSL_set_tlsext_host_name(st->ssl, server_name_.c_str());
+      SSL_set1_host(st->ssl, server_name_.c_str());
+    }
+    if (SSL_connect(st->ssl) != 1) {
+      std::string e = ssl_err();
+      st->close_free();
+      throw std::runtime_error("TLS handshake failed: " + e);
+    }
+    return st;
+  }
+
+  void release_tls_conn(const std::string& key, std::unique_ptr<Stream> st) {
+    std::lock_guard<std::mutex> lk(tls_mu_);
+    auto& pool = tls_pool_[key];
+    if (pool.size() < 8) {
+      pool.push_back(std::move(st));
+    } else {
+      st->close_free();
+    }
+  }
+
   void close_all() {
     std::unordered_map<std::string, std::shared_ptr<Conn>> conns;
     {
@@ -407,6 +628,10 @@ class XferClient {
       if (conn->reader.joinable()) conn->reader.join();
       ::close(conn->fd);
     }
+    std::lock_guard<std::mutex> lk2(tls_mu_);
+    for (auto& [key, pool] : tls_pool_)
+      for (auto& st : pool) st->close_free();
+    tls_pool_.clear();
   }
 
  private:
@@ -515,9 +740,14 @@ class XferClient {
   }
 
   std::string job_;
+  std::string server_name_;
+  SSL_CTX* ssl_ctx_ = nullptr;
   std::mutex conns_mu_;
   std::unordered_map<std::string, std::shared_ptr<Conn>> conns_;
   std::vector<std::shared_ptr<Conn>> dead_;  // kept until close_all
+  std::mutex tls_mu_;
+  std::unordered_map<std::string, std::vector<std::unique_ptr<Stream>>> tls_pool_;
+  std::atomic<uint64_t> tls_req_id_{1};
 };
 
 }  // namespace
@@ -525,7 +755,9 @@ class XferClient {
 PYBIND11_MODULE(_xfer, m) {
   m.doc() = "rayfed_amd C++ transport core (plaintext cross-silo hot path)";
   py::class_<XferServer>(m, "XferServer")
-      .def(py::init<int, std::string>(), py::arg("port"), py::arg("job_name"))
+      .def(py::init<int, std::string, std::string, std::string, std::string>(),
+           py::arg("port"), py::arg("job_name"), py::arg("tls_cert") = "",
+           py::arg("tls_key") = "", py::arg("tls_ca") = "")
       .def("start", &XferServer::start, py::arg("consume_cb"))
       .def("stop", &XferServer::stop,
            py::call_guard<py::gil_scoped_release>())
@@ -536,7 +768,11 @@ PYBIND11_MODULE(_xfer, m) {
       .def_property_readonly("received_op_count",
                              &XferServer::received_op_count);
   py::class_<XferClient>(m, "XferClient")
-      .def(py::init<std::string>(), py::arg("job_name"))
+      .def(py::init<std::string, std::string, std::string, std::string,
+                    std::string>(),
+           py::arg("job_name"), py::arg("tls_ca") = "",
+           py::arg("tls_cert") = "", py::arg("tls_key") = "",
+           py::arg("server_name") = "")
       .def(
           "send",
           [](XferClient& c, const std::string& host, int port,

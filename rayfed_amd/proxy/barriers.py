@@ -271,15 +271,13 @@ def _get_io_loop() -> IoLoop:
 
 
 def _use_cpp_transport(tls_config) -> bool:
-    """Default to the C++ transport core for plaintext jobs when built.
-    TLS rides the Python asyncio transport (ssl-integrated); RAYFED_TRANSPORT
-    ∈ {cpp, asyncio} forces a choice."""
+    """Default to the C++ transport core (plaintext AND TLS — OpenSSL with
+    mutual auth in csrc/xfer_core.cpp) when the extension is built;
+    RAYFED_TRANSPORT ∈ {cpp, asyncio} forces a choice."""
     import os
 
     mode = os.environ.get("RAYFED_TRANSPORT", "auto")
     if mode == "asyncio":
-        return False
-    if tls_config:
         return False
     from rayfed_amd.proxy.xfer import xfer_available
 
@@ -305,7 +303,7 @@ def start_receiver_proxy(
         from rayfed_amd.proxy.xfer import XferReceiverService
 
         service = XferReceiverService(
-            addresses[party], party, job_name, proxy_config
+            addresses[party], party, job_name, proxy_config, tls_config
         )
         _receiver_service = service
         _service_registry[receiver_proxy_name(job_name, use_global_proxy)] = service
@@ -339,7 +337,9 @@ def start_sender_proxy(
     if proxy_cls is None and _use_cpp_transport(tls_config):
         from rayfed_amd.proxy.xfer import XferSenderService
 
-        service = XferSenderService(addresses, party, job_name, proxy_config)
+        service = XferSenderService(
+            addresses, party, job_name, proxy_config, tls_config
+        )
         _sender_service = service
         _service_registry[sender_proxy_name(job_name, use_global_proxy)] = service
         return service

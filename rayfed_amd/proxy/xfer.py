@@ -81,18 +81,27 @@ class XferSenderService:
     """send() → concurrent Future; encode + C++ socket I/O on pool threads."""
 
     def __init__(self, addresses: Dict, party: str, job_name: str,
-                 proxy_config=None):
+                 proxy_config=None, tls_config=None):
         proxy_config = _coerce_config(proxy_config)
         self._addresses = addresses
         self._party = party
         self._job_name = job_name
         self._proxy_config = proxy_config
         xfer = _load_xfer()
+        tls_kw = {}
+        if tls_config:
+            tls_kw = {
+                "tls_ca": tls_config.get("ca_cert", ""),
+                "tls_cert": tls_config.get("cert", ""),
+                "tls_key": tls_config.get("key", ""),
+                "server_name": tls_config.get("target_name_override",
+                                              "localhost"),
+            }
         # Two connections per destination: control frames must not queue
         # behind a multi-GiB defer-ack consume on the bulk lane (a float
         # broadcast measured 100 ms stuck behind a 16 GB frame's consume).
-        self._client_ctl = xfer.XferClient(job_name)
-        self._client_bulk = xfer.XferClient(job_name)
+        self._client_ctl = xfer.XferClient(job_name, **tls_kw)
+        self._client_bulk = xfer.XferClient(job_name, **tls_kw)
         self._pool = ThreadPoolExecutor(max_workers=8, thread_name_prefix="xfer-send")
         self._retry = _Retry(
             getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
@@ -250,7 +259,7 @@ class XferReceiverService:
     """get_data() → concurrent Future; C++ mailbox, Python deserialization."""
 
     def __init__(self, listening_address: str, party: str, job_name: str,
-                 proxy_config=None):
+                 proxy_config=None, tls_config=None):
         proxy_config = _coerce_config(proxy_config)
         self._party = party
         self._job_name = job_name
@@ -260,8 +269,15 @@ class XferReceiverService:
         )
         port = int(listening_address.rsplit(":", 1)[1])
         xfer = _load_xfer()
+        tls_kw = {}
+        if tls_config:
+            tls_kw = {
+                "tls_cert": tls_config.get("cert", ""),
+                "tls_key": tls_config.get("key", ""),
+                "tls_ca": tls_config.get("ca_cert", ""),
+            }
         try:
-            self._server = xfer.XferServer(port, job_name)
+            self._server = xfer.XferServer(port, job_name, **tls_kw)
             self._server.start(self._consume_deferred)
         except RuntimeError as e:
             raise AssertionError(

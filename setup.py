@@ -31,6 +31,7 @@ setup(
             name="rayfed_amd._xfer",
             sources=["csrc/xfer_core.cpp"],
             extra_compile_args=["-O3", "-std=c++17", "-pthread"],
+            libraries=["ssl", "crypto"],
         ),
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
