@@ -49,3 +49,38 @@ def test_dict2tuple():
     assert dict2tuple(None) == []
     assert sorted(dict2tuple({"a": 1, "b": 2})) == [("a", 1), ("b", 2)]
     assert dict2tuple([("x", 3)]) == [("x", 3)]
+
+
+def test_start_command():
+    from rayfed_amd.utils import start_command
+
+    assert start_command("echo hello").strip() == "hello"
+    import pytest as _pytest
+
+    with _pytest.raises(RuntimeError):
+        start_command("echo oops 1>&2")
+
+
+def test_materialize_resolves_nested_refs():
+    from rayfed_amd.runtime.object_ref import ObjectRef
+    from rayfed_amd.utils import materialize
+
+    tree = {"a": [ObjectRef.from_value(1), 2], "b": (ObjectRef.from_value(3),)}
+    assert materialize(tree) == {"a": [1, 2], "b": (3,)}
+
+
+def test_is_cython_false_for_plain_function():
+    from rayfed_amd.utils import is_cython
+
+    assert is_cython(lambda: None) is False
+
+
+def test_setup_logger_injects_party_fields(capsys):
+    import logging
+
+    from rayfed_amd.utils import setup_logger
+
+    setup_logger(logging_level="info", party="alice", job_name="jobx")
+    logging.getLogger("rayfed_amd.test").info("hello-log")
+    err = capsys.readouterr().err
+    assert "[alice]" in err and "[jobx]" in err and "hello-log" in err
