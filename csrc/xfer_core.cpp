@@ -53,6 +53,9 @@ namespace py = pybind11;
 namespace {
 
 constexpr uint8_t kFlagDeferAck = 1;
+// Reserved seq id of the init-time readiness barrier (constants.PING_SEQ_ID):
+// acked without parking so repeated pings never leak mailbox slots.
+constexpr const char* kPingSeqId = "ping";
 
 // ---------------------------------------------------------------- utilities
 static void write_all(int fd, const char* data, size_t n) {
@@ -215,8 +218,12 @@ class XferServer {
     if (ssl_ctx_) SSL_CTX_free(ssl_ctx_);
   }
 
-  // consume_cb(body_bytes) -> int code; called (with the GIL) for
-  // DEFER_ACK frames only.  Delivery happens before the ack is written.
+  // consume_cb(up, down, body_bytes, token) is called (with the GIL) for
+  // DEFER_ACK frames only.  It must NOT block: it hands the decode to a
+  // Python worker thread and returns immediately; the worker acks by calling
+  // complete(token, code, result) when the consume (H2D + CRC) is done.
+  // This keeps a multi-GiB consume from stalling the connection thread —
+  // other frames on the same connection keep flowing while it runs.
   void start(py::object consume_cb) {
     consume_cb_ = std::move(consume_cb);
     listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
@@ -251,7 +258,7 @@ class XferServer {
     ::close(listen_fd_);
     {
       std::lock_guard<std::mutex> lk(conn_mu_);
-      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+      for (auto& c : conns_) ::shutdown(c->st.fd, SHUT_RDWR);
     }
     if (accept_thread_.joinable()) accept_thread_.join();
     {
@@ -260,10 +267,36 @@ class XferServer {
         if (t.joinable()) t.join();
       conn_threads_.clear();
     }
+    {  // outstanding deferred acks become no-ops
+      std::lock_guard<std::mutex> lk(tok_mu_);
+      tokens_.clear();
+    }
     // Unblock any waiting get_data.
     std::lock_guard<std::mutex> lk(mail_mu_);
     stopped_ = true;
     mail_cv_.notify_all();
+  }
+
+  // Ack a deferred-consume frame from a Python worker thread (GIL released
+  // by the binding).  No-op if the connection died or the server stopped.
+  void complete(uint64_t token, int code, const std::string& result) {
+    std::shared_ptr<SrvConn> conn;
+    uint64_t req_id = 0;
+    {
+      std::lock_guard<std::mutex> lk(tok_mu_);
+      auto it = tokens_.find(token);
+      if (it == tokens_.end()) return;
+      conn = it->second.first;
+      req_id = it->second.second;
+      tokens_.erase(it);
+    }
+    if (!conn->alive.load()) return;
+    try {
+      write_ack(*conn, req_id, (uint16_t)code, result);
+    } catch (...) {
+      conn->alive = false;
+      ::shutdown(conn->st.fd, SHUT_RDWR);
+    }
   }
 
   // Blocking fetch (GIL released by the binding); returns body bytes.
@@ -319,6 +352,30 @@ class XferServer {
   uint64_t received_op_count() const { return recv_count_.load(); }
 
  private:
+  // A server-side connection: the conn thread reads frames; acks may be
+  // written by the conn thread (inline path) or a Python worker completing
+  // a deferred consume — write_mu keeps them from interleaving.
+  struct SrvConn {
+    Stream st;
+    std::mutex write_mu;
+    std::atomic<bool> alive{true};
+  };
+
+  static void write_ack(SrvConn& c, uint64_t req_id, uint16_t code,
+                        const std::string& result) {
+    char ack[14];
+    uint32_t ack_len = 10 + (uint32_t)result.size();
+    memcpy(ack, &ack_len, 4);
+    memcpy(ack + 4, &req_id, 8);
+    memcpy(ack + 12, &code, 2);
+    std::lock_guard<std::mutex> lk(c.write_mu);
+    // Re-check under the lock: teardown closes/frees the stream while
+    // holding write_mu, so a dead conn is never written (no SSL* UAF).
+    if (!c.alive.load()) throw std::runtime_error("connection closed");
+    c.st.write_all(ack, 14);
+    if (!result.empty()) c.st.write_all(result.data(), result.size());
+  }
+
   void accept_loop() {
     while (running_) {
       int fd = ::accept(listen_fd_, nullptr, nullptr);
@@ -327,20 +384,21 @@ class XferServer {
         break;
       }
       set_sock_opts(fd);
+      auto conn = std::make_shared<SrvConn>();
+      conn->st.fd = fd;
       std::lock_guard<std::mutex> lk(conn_mu_);
-      conn_fds_.push_back(fd);
-      conn_threads_.emplace_back([this, fd] { conn_loop(fd); });
-
+      conns_.push_back(conn);
+      conn_threads_.emplace_back([this, conn] { conn_loop(conn); });
     }
   }
 
-  void conn_loop(int fd) {
-    Stream st;
-    st.fd = fd;
+  void conn_loop(std::shared_ptr<SrvConn> conn) {
+    Stream& st = conn->st;
     if (ssl_ctx_) {
       st.ssl = SSL_new(ssl_ctx_);
-      SSL_set_fd(st.ssl, fd);
+      SSL_set_fd(st.ssl, st.fd);
       if (SSL_accept(st.ssl) != 1) {
+        conn->alive = false;
         st.close_free();
         return;
       }
@@ -373,16 +431,32 @@ class XferServer {
       if (job != job_) {
         code = 417;
         result = "JobName mis-match: expected " + job_ + ", got " + job;
+      } else if (up == kPingSeqId && down == kPingSeqId) {
+        // Readiness ping: ack without parking (nothing ever consumes it)
+        // and without counting it as a received data op.
       } else if (flags & kFlagDeferAck) {
-        // shm-lane frame: Python consumes (H2D + CRC) before we ack.
-        py::gil_scoped_acquire gil;
-        try {
-          py::bytes body(buf.data() + off, buf.size() - off);
-          code = (uint16_t)py::cast<int>(consume_cb_(up, down, body));
-        } catch (const std::exception& e) {
-          code = 500;
-          result = std::string("consume failed: ") + e.what();
+        // shm/IPC-lane frame: Python consumes (H2D + CRC) before the ack —
+        // hand it off and keep reading; the worker acks via complete().
+        uint64_t token = next_token_.fetch_add(1);
+        {
+          std::lock_guard<std::mutex> lk(tok_mu_);
+          tokens_[token] = {conn, req_id};
         }
+        bool handed_off = false;
+        {
+          py::gil_scoped_acquire gil;
+          try {
+            py::bytes body(buf.data() + off, buf.size() - off);
+            consume_cb_(up, down, body, token);
+            handed_off = true;
+          } catch (const std::exception& e) {
+            code = 500;
+            result = std::string("consume dispatch failed: ") + e.what();
+          }
+        }
+        if (handed_off) continue;  // ack comes later from complete()
+        std::lock_guard<std::mutex> lk(tok_mu_);
+        tokens_.erase(token);
       } else {
         recv_count_.fetch_add(1);
         std::lock_guard<std::mutex> lk(mail_mu_);
@@ -391,19 +465,17 @@ class XferServer {
         mail_cv_.notify_all();
       }
 
-      char ack[14];
-      uint32_t ack_len = 10 + (uint32_t)result.size();
-      memcpy(ack, &ack_len, 4);
-      memcpy(ack + 4, &req_id, 8);
-      memcpy(ack + 12, &code, 2);
       try {
-        st.write_all(ack, 14);
-        if (!result.empty()) st.write_all(result.data(), result.size());
+        write_ack(*conn, req_id, code, result);
       } catch (...) {
         break;
       }
     }
-    st.close_free();
+    {
+      std::lock_guard<std::mutex> lk(conn->write_mu);
+      conn->alive = false;
+      st.close_free();
+    }
   }
 
   std::string job_;
@@ -413,12 +485,16 @@ class XferServer {
   bool stopped_ = false;
   std::thread accept_thread_;
   std::mutex conn_mu_;
-  std::vector<int> conn_fds_;
+  std::vector<std::shared_ptr<SrvConn>> conns_;
   std::vector<std::thread> conn_threads_;
   std::mutex mail_mu_;
   std::condition_variable mail_cv_;
   std::map<std::string, std::string> mail_;
   std::atomic<uint64_t> recv_count_{0};
+  std::mutex tok_mu_;
+  std::unordered_map<uint64_t, std::pair<std::shared_ptr<SrvConn>, uint64_t>>
+      tokens_;
+  std::atomic<uint64_t> next_token_{1};
   py::object consume_cb_;
   SSL_CTX* ssl_ctx_ = nullptr;
 };
@@ -595,10 +671,11 @@ class XferClient {
     st->fd = fd;
     st->ssl = SSL_new(ssl_ctx_);
     SSL_set_fd(st->ssl, fd);
-    if (!server_name_.empty()) {
-      SSL_set_tlsext_host_name(st->ssl, server_name_.c_str());
-      SSL_set1_host(st->ssl, server_name_.c_str());
-    }
+    // Verify against the destination host unless a name is pinned explicitly
+    // (grpc.ssl_target_name_override semantics) — never a fixed default.
+    const std::string& sni = server_name_.empty() ? host : server_name_;
+    SSL_set_tlsext_host_name(st->ssl, sni.c_str());
+    SSL_set1_host(st->ssl, sni.c_str());
     if (SSL_connect(st->ssl) != 1) {
       std::string e = ssl_err();
       st->close_free();
@@ -763,6 +840,9 @@ PYBIND11_MODULE(_xfer, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("get_data", &XferServer::get_data, py::arg("up"), py::arg("down"),
            py::arg("timeout_s") = 600.0)
+      .def("complete", &XferServer::complete, py::arg("token"),
+           py::arg("code"), py::arg("result") = std::string(),
+           py::call_guard<py::gil_scoped_release>())
       .def("post", &XferServer::post, py::call_guard<py::gil_scoped_release>())
       .def("try_take", &XferServer::try_take)
       .def_property_readonly("received_op_count",

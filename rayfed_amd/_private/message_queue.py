@@ -60,6 +60,11 @@ class MessageQueueManager:
     def is_started(self) -> bool:
         return self._thread is not None and self._thread.is_alive()
 
+    def abandon(self) -> None:
+        """Drop whatever is still queued without joining the thread — safe to
+        call from the queue's own handler (``stop(False)`` is not: it joins)."""
+        self._abandon.set()
+
     def stop(self, wait_for_sending: bool = True) -> None:
         """Stop the polling thread.
 

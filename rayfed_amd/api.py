@@ -77,6 +77,17 @@ def init(
     )
     gpu_plane_dict = config.get("gpu_data_plane", {})
 
+    # Placement labels steer Ray actor scheduling in the reference
+    # (barriers.py:264-271); this engine hosts the proxies in-process, so
+    # there is nothing to place.  Warn loudly instead of ignoring silently.
+    for label_knob in ("send_resource_label", "recv_resource_label"):
+        if getattr(cross_silo_comm_config, label_knob, None):
+            logger.warning(
+                "cross_silo_comm.%s is not applicable: proxies run "
+                "in-process (no actor placement); the label is ignored.",
+                label_knob,
+            )
+
     ctx = init_global_context(
         current_party=party,
         job_name=job_name,

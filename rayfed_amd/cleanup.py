@@ -139,7 +139,7 @@ class CleanupManager:
                 if not self.continue_waiting_for_data_sending_on_error:
                     # Abandon the remaining queued sends (asynchronously: the
                     # queue thread itself must not join itself).
-                    self._sending_data_q._abandon.set()
+                    self._sending_data_q.abandon()
             return False
 
     def _process_error_sending_task_return(self, item: _PendingSend) -> bool:

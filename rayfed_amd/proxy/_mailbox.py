@@ -10,7 +10,7 @@ from __future__ import annotations
 import asyncio
 from typing import Dict, Optional, Tuple
 
-from rayfed_amd._private import serialization
+from rayfed_amd._private import constants, serialization
 from rayfed_amd.ops import tensor_codec
 from rayfed_amd.proxy.grpc import frames
 
@@ -54,6 +54,14 @@ class Mailbox:
         bad = self.check_job(header)
         if bad is not None:
             return bad
+        if (
+            header.get("up") == constants.PING_SEQ_ID
+            and header.get("down") == constants.PING_SEQ_ID
+        ):
+            # Readiness ping: ack without parking — nothing ever consumes
+            # these, so parking would leak a slot per ping — and without
+            # counting it as a received data op.
+            return 200, "OK"
         self._park(header, (kind, header, payload))
         return 200, "OK"
 

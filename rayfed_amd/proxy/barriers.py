@@ -270,7 +270,7 @@ def _get_io_loop() -> IoLoop:
     return _io_loop
 
 
-def _use_cpp_transport(tls_config) -> bool:
+def _use_cpp_transport() -> bool:
     """Default to the C++ transport core (plaintext AND TLS — OpenSSL with
     mutual auth in csrc/xfer_core.cpp) when the extension is built;
     RAYFED_TRANSPORT ∈ {cpp, asyncio} forces a choice."""
@@ -299,7 +299,7 @@ def start_receiver_proxy(
     use_global_proxy: bool = True,
 ):
     global _receiver_service
-    if proxy_cls is None and _use_cpp_transport(tls_config):
+    if proxy_cls is None and _use_cpp_transport():
         from rayfed_amd.proxy.xfer import XferReceiverService
 
         service = XferReceiverService(
@@ -334,7 +334,7 @@ def start_sender_proxy(
     use_global_proxy: bool = True,
 ):
     global _sender_service
-    if proxy_cls is None and _use_cpp_transport(tls_config):
+    if proxy_cls is None and _use_cpp_transport():
         from rayfed_amd.proxy.xfer import XferSenderService
 
         service = XferSenderService(

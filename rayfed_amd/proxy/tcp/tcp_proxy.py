@@ -63,9 +63,16 @@ def _parse_duration_s(v, default: float) -> float:
 
 
 class _RetryPolicy:
-    def __init__(self, d: Optional[dict]):
+    def __init__(self, d: Optional[dict], proxy_max_restarts: Optional[int] = None):
         d = d or {}
-        self.max_attempts = int(d.get("maxAttempts", 5))
+        # proxy_max_restarts restarted the sender proxy ACTOR in the
+        # reference (barriers.py:301-307); here the proxy is in-process, so
+        # it maps to the reconnect budget per send: N restarts → N+1
+        # attempts.  An explicit retry policy's maxAttempts wins.
+        if "maxAttempts" in d or proxy_max_restarts is None:
+            self.max_attempts = int(d.get("maxAttempts", 5))
+        else:
+            self.max_attempts = int(proxy_max_restarts) + 1
         self.initial_backoff = _parse_duration_s(d.get("initialBackoff"), 5.0)
         self.max_backoff = _parse_duration_s(d.get("maxBackoff"), 30.0)
         self.multiplier = float(d.get("backoffMultiplier", 2))
@@ -98,6 +105,11 @@ class _Connection:
         self.pending: Dict[int, asyncio.Future] = {}
         self.ids = itertools.count(1)
         self.alive = True
+        # Held across ALL writes of one frame: the chunk loop below awaits
+        # drain() mid-frame, and without the lock a concurrent send coroutine
+        # pipelining on this connection could interleave its bytes inside the
+        # first request's frame, corrupting the length-prefixed stream.
+        self.write_lock = asyncio.Lock()
         self.reader_task = asyncio.get_running_loop().create_task(self._read_loop())
 
     async def _read_loop(self):
@@ -129,21 +141,22 @@ class _Connection:
         total = len(prefix) + sum(len(p) for p in parts)
         head = (total + 8).to_bytes(8, "little") + req_id.to_bytes(8, "little")
         _CHUNK = 8 << 20  # bound transport buffering for multi-GiB parts
-        if total <= 65536:
-            # Small request: one write, one TCP segment.
-            self.writer.write(head + prefix + b"".join(bytes(p) for p in parts))
-        else:
-            self.writer.write(head)
-            self.writer.write(prefix)
-            for p in parts:
-                if len(p) <= _CHUNK:
-                    self.writer.write(p)
-                else:
-                    mv = memoryview(p)
-                    for off in range(0, len(mv), _CHUNK):
-                        self.writer.write(mv[off : off + _CHUNK])
-                        await self.writer.drain()
-        await self.writer.drain()
+        async with self.write_lock:  # one frame's bytes stay contiguous
+            if total <= 65536:
+                # Small request: one write, one TCP segment.
+                self.writer.write(head + prefix + b"".join(bytes(p) for p in parts))
+            else:
+                self.writer.write(head)
+                self.writer.write(prefix)
+                for p in parts:
+                    if len(p) <= _CHUNK:
+                        self.writer.write(p)
+                    else:
+                        mv = memoryview(p)
+                        for off in range(0, len(mv), _CHUNK):
+                            self.writer.write(mv[off : off + _CHUNK])
+                            await self.writer.drain()
+            await self.writer.drain()
         return await asyncio.wait_for(fut, timeout=timeout)
 
     async def close(self):
@@ -169,7 +182,10 @@ class TcpSenderProxy(base_proxy.SenderProxy):
         self.gpu_plane = None
         self.last_sent_bytes = 0
         self._retry = _RetryPolicy(
-            getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None
+            getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None,
+            proxy_max_restarts=(
+                proxy_config.proxy_max_restarts if proxy_config else None
+            ),
         )
         self._timeout_s = (
             (proxy_config.timeout_in_ms / 1000.0)
@@ -177,8 +193,13 @@ class TcpSenderProxy(base_proxy.SenderProxy):
             else 60.0
         )
         self._ssl = _client_ssl_context(tls_config) if tls_config else None
+        # Verify the certificate against the destination host unless the user
+        # pins a name explicitly (same semantics as the reference's
+        # grpc.ssl_target_name_override, grpc_proxy.py:131-136) — a hard
+        # 'localhost' default would let any cert with a localhost SAN
+        # authenticate as any party on any host.
         self._server_hostname = (
-            tls_config.get("target_name_override", "localhost") if tls_config else None
+            tls_config.get("target_name_override") if tls_config else None
         )
 
     async def _ensure_conn(self, dest_party: str) -> _Connection:
@@ -197,7 +218,7 @@ class TcpSenderProxy(base_proxy.SenderProxy):
                 host,
                 int(port),
                 ssl=self._ssl,
-                server_hostname=self._server_hostname if self._ssl else None,
+                server_hostname=(self._server_hostname or host) if self._ssl else None,
                 limit=16 << 20,
             )
             writer.transport.set_write_buffer_limits(high=1 << 26)
