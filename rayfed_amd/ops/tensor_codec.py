@@ -106,6 +106,47 @@ def _tensor_bytes_cpu(t: "torch.Tensor"):
     return t.view(-1).view(torch.uint8).numpy()
 
 
+def route_for(t: "torch.Tensor", gpu_plane=None, shm: bool = False) -> str:
+    """Which lane one tensor rides — see :func:`route_for_spec`."""
+    return route_for_spec(
+        t.numel() * t.element_size(),
+        t.dtype,
+        t.device.type,
+        gpu_plane,
+        shm,
+    )
+
+
+def route_for_spec(
+    nbytes: int, dtype, device_type: str, gpu_plane=None, shm: bool = False
+) -> str:
+    """Which lane a tensor of this (size, dtype, device) rides: payload
+    (inline pickle bytes), shm_cpu, shm_gpu, shm_chunked, ipc (multi-slab
+    device-IPC), or ipc_group (arena slab).  Exposed so benchmarks and
+    diagnostics can report the lane without packing anything."""
+    import os as _os
+
+    if not shm:
+        return "payload"
+    from rayfed_amd.ops import shm_pool
+
+    if nbytes < shm_pool.SHM_MIN_BYTES:
+        return "payload"
+    if gpu_plane is None or device_type != "cuda":
+        return "shm_cpu"
+    ipc_on = _os.environ.get("RAYFED_IPC", "1") != "0"
+    wire_fp8 = (
+        gpu_plane.config.wire_dtype == "fp8e4m3" and dtype == torch.bfloat16
+    )
+    if ipc_on and not wire_fp8 and nbytes <= gpu_plane.IPC_SLAB_BYTES:
+        return "ipc_group"
+    if ipc_on:
+        return "ipc"
+    if not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
+        return "shm_chunked"
+    return "shm_gpu"
+
+
 def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memoryview]]:
     """Serialize ``obj``; returns (header_extras, payload_parts).
 
@@ -129,13 +170,6 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
     # Route every tensor first so mid-size device tensors can be ARENA-packed
     # together into shared IPC slabs (a state_dict push ships ~16 slab
     # handles instead of one per tensor).
-    import os as _os
-
-    ipc_on = _os.environ.get("RAYFED_IPC", "1") != "0"
-    if shm:
-        from rayfed_amd.ops import shm_pool
-
-        shm_min = shm_pool.SHM_MIN_BYTES
     routes = []
     for t in pickler.tensors:
         if torch is None:
@@ -143,25 +177,7 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
         dtype = _DTYPE_TO_STR.get(t.dtype)
         if dtype is None:
             raise TypeError(f"unsupported tensor dtype {t.dtype}")
-        nbytes = t.numel() * t.element_size()
-        route = "payload"
-        if shm and nbytes >= shm_min:
-            if gpu_plane is not None and t.device.type == "cuda":
-                wire_fp8 = (
-                    gpu_plane.config.wire_dtype == "fp8e4m3"
-                    and t.dtype == torch.bfloat16
-                )
-                if ipc_on and not wire_fp8 and nbytes <= gpu_plane.IPC_SLAB_BYTES:
-                    route = "ipc_group"
-                elif ipc_on:
-                    route = "ipc"
-                elif not wire_fp8 and nbytes >= 2 * gpu_plane.config.chunk_bytes:
-                    route = "shm_chunked"
-                else:
-                    route = "shm_gpu"
-            else:
-                route = "shm_cpu"
-        routes.append(route)
+        routes.append(route_for(t, gpu_plane, shm))
 
     group_members = [i for i, r in enumerate(routes) if r == "ipc_group"]
     if len(group_members) == 1:
