@@ -112,6 +112,17 @@ static void set_sock_opts(int fd) {
   setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof(buf));
 }
 
+// Bounded busy-wait window before parking on a condition variable; tunable
+// via RAYFED_SPIN_US (0 disables).  Applied on the tiny-message hot paths
+// only (mailbox fetch, ack wait) — bulk transfers park immediately.
+static int spin_us() {
+  static int v = [] {
+    const char* e = getenv("RAYFED_SPIN_US");
+    return e ? atoi(e) : 120;
+  }();
+  return v;
+}
+
 static std::string ssl_err() {
   char buf[256];
   ERR_error_string_n(ERR_get_error(), buf, sizeof(buf));
@@ -300,25 +311,49 @@ class XferServer {
   }
 
   // Blocking fetch (GIL released by the binding); returns body bytes.
+  // Spins briefly before the cv wait: on the tiny-task hot path the reply
+  // lands within ~100 us, and catching it in the spin saves a futex
+  // wake-from-idle (~50 us when the waiter has gone cold).
   py::bytes get_data(const std::string& up, const std::string& down,
                      double timeout_s) {
     std::string key = up + '\x00' + down;
     std::string body;
     {
       py::gil_scoped_release release;
-      std::unique_lock<std::mutex> lk(mail_mu_);
       auto deadline = std::chrono::steady_clock::now() +
                       std::chrono::duration<double>(timeout_s);
-      while (true) {
-        auto it = mail_.find(key);
-        if (it != mail_.end()) {
-          body = std::move(it->second);
-          mail_.erase(it);
-          break;
+      const auto spin_until = std::chrono::steady_clock::now() +
+                              std::chrono::microseconds(spin_us());
+      bool got = false;
+      while (!got && std::chrono::steady_clock::now() < spin_until) {
+        {
+          std::unique_lock<std::mutex> lk(mail_mu_, std::try_to_lock);
+          if (lk.owns_lock()) {
+            auto it = mail_.find(key);
+            if (it != mail_.end()) {
+              body = std::move(it->second);
+              mail_.erase(it);
+              got = true;
+              break;
+            }
+            if (stopped_) throw std::runtime_error("server stopped");
+          }
         }
-        if (stopped_) throw std::runtime_error("server stopped");
-        if (mail_cv_.wait_until(lk, deadline) == std::cv_status::timeout)
-          throw std::runtime_error("get_data timeout");
+        for (int i = 0; i < 32; ++i) __builtin_ia32_pause();
+      }
+      if (!got) {
+        std::unique_lock<std::mutex> lk(mail_mu_);
+        while (true) {
+          auto it = mail_.find(key);
+          if (it != mail_.end()) {
+            body = std::move(it->second);
+            mail_.erase(it);
+            break;
+          }
+          if (stopped_) throw std::runtime_error("server stopped");
+          if (mail_cv_.wait_until(lk, deadline) == std::cv_status::timeout)
+            throw std::runtime_error("get_data timeout");
+        }
       }
     }
     return py::bytes(body);
@@ -580,6 +615,13 @@ class XferClient {
           throw;
         }
       }
+      // Spin briefly for the ack (loopback RTT ~10-50 us) before parking.
+      const auto spin_until = std::chrono::steady_clock::now() +
+                              std::chrono::microseconds(spin_us());
+      while (!pending->done.load(std::memory_order_acquire) &&
+             std::chrono::steady_clock::now() < spin_until) {
+        for (int i = 0; i < 32; ++i) __builtin_ia32_pause();
+      }
       std::unique_lock<std::mutex> lk(pending->mu);
       auto deadline = std::chrono::steady_clock::now() +
                       std::chrono::duration<double>(timeout_s);
@@ -715,7 +757,7 @@ class XferClient {
   struct Pending {
     std::mutex mu;
     std::condition_variable cv;
-    bool done = false;
+    std::atomic<bool> done{false};  // atomic: spun on outside the lock
     bool broken = false;
     uint16_t code = 0;
     std::string result;

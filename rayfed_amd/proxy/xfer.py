@@ -18,6 +18,7 @@ import logging
 import threading
 import time
 from concurrent.futures import Future, ThreadPoolExecutor
+from concurrent.futures import TimeoutError as FutureTimeoutError
 from typing import Dict, Optional
 
 from rayfed_amd import config as fed_config
@@ -400,14 +401,21 @@ class XferReceiverService:
                 fut.set_exception(e)
             return fut
 
-        def _wait():
+        def _wait(timeout=None):
             # Parity with the reference's recv semantics: wait indefinitely
             # for the peer's push (failure paths deliver an error object on
             # the same seq ids instead of leaving this hanging).  Slice the
             # C++ wait so server stop still unblocks promptly.
+            deadline = None if timeout is None else time.monotonic() + timeout
             while True:
+                if deadline is None:
+                    chunk = 60.0
+                else:
+                    chunk = min(60.0, deadline - time.monotonic())
+                    if chunk <= 0:
+                        raise FutureTimeoutError()
                 try:
-                    b = self._server.get_data(up, down, 60.0)
+                    b = self._server.get_data(up, down, chunk)
                     break
                 except RuntimeError as e:
                     if "timeout" in str(e):
@@ -415,7 +423,10 @@ class XferReceiverService:
                     raise  # server stopped
             return self._take(up, down, b)
 
-        return self._pool.submit(_wait)
+        # Lazy: the consumer's own thread performs the C++ mailbox wait on
+        # result() — no recv-pool hop (one futex wake-from-idle fewer per
+        # cross-party edge on the tiny-task critical path).
+        return _LazyFuture(_wait)
 
     def _get_stats(self) -> Dict[str, int]:
         return {"receive_op_count": self.receive_op_count}
@@ -424,3 +435,39 @@ class XferReceiverService:
         self._server.stop()
         self._pool.shutdown(wait=False, cancel_futures=True)
         self._consume_pool.shutdown(wait=False, cancel_futures=True)
+
+
+class _LazyFuture(Future):
+    """A Future whose value is produced by the first result() caller running
+    ``fetch(timeout)`` in its own thread; other waiters block on the Future
+    proper.  ``fetch`` raising concurrent.futures.TimeoutError leaves the
+    future unresolved (standard result(timeout) semantics)."""
+
+    def __init__(self, fetch):
+        super().__init__()
+        self._fetch = fetch
+        self._claim = threading.Lock()
+
+    def result(self, timeout=None):
+        if not self.done() and self._claim.acquire(blocking=False):
+            try:
+                if not self.done():
+                    try:
+                        value = self._fetch(timeout)
+                    except FutureTimeoutError:
+                        raise  # not an outcome: the caller just gave up
+                    except BaseException as e:  # noqa: BLE001
+                        self.set_exception(e)
+                    else:
+                        self.set_result(value)
+            finally:
+                self._claim.release()
+        return super().result(timeout)
+
+    def exception(self, timeout=None):
+        if not self.done():
+            try:
+                self.result(timeout)
+            except BaseException:  # noqa: BLE001
+                pass
+        return super().exception(timeout)

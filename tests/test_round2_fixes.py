@@ -143,6 +143,9 @@ def test_concurrent_shm_deferred_consumes(loop_env):
     from rayfed_amd.ops import shm_pool
 
     shm_pool.detach_all()
+    # Drain the sender pool too — later tests assert exact segment reuse and
+    # must not inherit this test's 8 MiB segments.
+    shm_pool.get_send_pool().shutdown()
 
 
 # ---------------------------------------------------------------- knob tests
@@ -180,13 +183,22 @@ def test_max_concurrency_bounds_xfer_pools():
         recv.stop()
 
 
-def test_resource_labels_warn(caplog):
+def test_resource_labels_warn():
     import logging
 
     import rayfed_amd as fed
 
+    records = []
+
+    class _Capture(logging.Handler):
+        def emit(self, record):
+            records.append(record.getMessage())
+
+    api_logger = logging.getLogger("rayfed_amd.api")
+    handler = _Capture(level=logging.WARNING)
+    api_logger.addHandler(handler)
     addrs = make_addresses(["alice"])
-    with caplog.at_level(logging.WARNING, logger="rayfed_amd.api"):
+    try:
         fed.init(
             addresses=addrs,
             party="alice",
@@ -197,10 +209,10 @@ def test_resource_labels_warn(caplog):
                 }
             },
         )
-    try:
-        text = caplog.text
+        text = "\n".join(records)
         assert "send_resource_label" in text
         assert "recv_resource_label" in text
         assert "ignored" in text
     finally:
+        api_logger.removeHandler(handler)
         fed.shutdown()
