@@ -123,6 +123,7 @@ class XferSenderService:
         # Two connections per destination: control frames must not queue
         # behind a multi-GiB defer-ack consume on the bulk lane (a float
         # broadcast measured 100 ms stuck behind a 16 GB frame's consume).
+        self._tls = bool(tls_config)
         self._client_ctl = xfer.XferClient(job_name, **tls_kw)
         self._client_bulk = xfer.XferClient(job_name, **tls_kw)
         workers = (
@@ -170,13 +171,152 @@ class XferSenderService:
         self._same_host_cache[dest_party] = same
         return same
 
+    # Frames at or below this ride the inline fast path: written by the
+    # thread that produced the data (send_async), acked lazily.
+    _INLINE_MAX = 256 << 10
+
     def send(self, dest_party, data, upstream_seq_id, downstream_seq_id) -> Future:
         with self._stats_lock:
             self.send_op_count += 1
-        return self._pool.submit(
-            self._send_blocking, dest_party, data,
-            str(upstream_seq_id), str(downstream_seq_id),
-        )
+        up = str(upstream_seq_id)
+        down = str(downstream_seq_id)
+        if self._tls:
+            # send_async is plaintext-only (an SSL* cannot take a concurrent
+            # reader+writer); TLS keeps the pooled blocking path.
+            return self._pool.submit(
+                self._send_blocking, dest_party, data, up, down
+            )
+
+        # Continuation-style send: issue the frame the moment the payload is
+        # ready, IN the producing thread (already hot) — a pool thread woken
+        # from idle costs 60-150 us, measured dominant on the tiny-task path.
+        state: dict = {}
+        ready = threading.Event()
+
+        def _issue(_f=None):
+            try:
+                d = data
+                if isinstance(d, ObjectRef):
+                    d = d.result()  # done (callback path); error propagates
+                body_parts, extras, defer_ack, nbytes = self._encode_frame(
+                    dest_party, d, up, down
+                )
+                if not defer_ack and nbytes <= self._INLINE_MAX:
+                    from rayfed_amd._private import tracing
+
+                    if tracing.enabled:
+                        tracing.event("xfer.send_inline", "xsilo",
+                                      dest=dest_party, up=up, down=down)
+                    host, port = self._addresses[dest_party].rsplit(":", 1)
+                    state["t0"] = time.perf_counter()
+                    state.update(
+                        parts=body_parts, extras=extras, nbytes=nbytes,
+                        host=host, port=int(port),
+                    )
+                    try:
+                        state["handle"] = self._client_ctl.send_async(
+                            host, int(port), up, down, body_parts, False
+                        )
+                    except RuntimeError:
+                        # e.g. peer not up yet: the pooled path carries the
+                        # retry/backoff budget (async-startup semantics).
+                        tensor_codec.release_parts(extras)
+                        state.clear()
+                        state["fut"] = self._pool.submit(
+                            self._send_blocking, dest_party, d, up, down
+                        )
+                else:
+                    state["fut"] = self._pool.submit(
+                        self._send_blocking, dest_party, d, up, down
+                    )
+            except BaseException as e:  # noqa: BLE001
+                state["exc"] = e
+            finally:
+                ready.set()
+
+        def _fetch(timeout=None):
+            if not ready.wait(timeout):
+                raise FutureTimeoutError()
+            if "exc" in state:
+                raise state["exc"]
+            if "fut" in state:
+                return state["fut"].result(timeout)
+            err = True
+            try:
+                try:
+                    code, result = self._client_ctl.wait_ack(
+                        state["handle"], self._timeout_s
+                    )
+                except RuntimeError:
+                    # Connection broke / ack lost: blocking re-send with the
+                    # retry budget (frames are idempotent — the mailbox is
+                    # keyed by seq ids).
+                    code, result = self._send_with_retry(
+                        self._client_ctl, state["host"], state["port"],
+                        up, down, state["parts"], False,
+                    )
+                if 400 <= code < 500:
+                    raise RuntimeError(
+                        f"[{code}] send to {dest_party} rejected: {result}"
+                    )
+                if code >= 500:
+                    raise RuntimeError(
+                        f"[{code}] send to {dest_party} failed: {result}"
+                    )
+                err = False
+                return True
+            finally:
+                tensor_codec.release_parts(state["extras"])
+                secs = time.perf_counter() - state["t0"]
+                with self._stats_lock:
+                    edge = self._edges.setdefault(dest_party, self._edge_cls())
+                    edge.record(state["nbytes"], secs, err)
+
+        if isinstance(data, ObjectRef) and not data.future.done():
+            data.future.add_done_callback(_issue)
+        else:
+            _issue()
+        return _LazyFuture(_fetch)
+
+    def _encode_frame(self, dest_party, data, up, down):
+        """Serialize one payload into wire parts.  Returns
+        (body_parts, extras, defer_ack, nbytes)."""
+        from rayfed_amd.exceptions import FedRemoteError
+        from rayfed_amd.ops import shm_pool
+
+        header = {"job": self._job_name, "up": up, "down": down}
+        if isinstance(data, FedRemoteError):
+            from rayfed_amd._private import serialization
+
+            extras = {"tensors": []}
+            body_parts = [
+                frames.encode_frame_prefix(frames.KIND_ERROR, header),
+                serialization.dumps(data),
+            ]
+            defer_ack = False
+        else:
+            use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
+            extras, parts = tensor_codec.encode(data, self.gpu_plane, use_shm)
+            defer_ack = False
+            if extras["tensors"]:
+                wire_header = {
+                    k: v for k, v in extras.items() if k != "_releases"
+                }
+                header.update(wire_header)
+                defer_ack = any(
+                    "shm" in m or "ipc_slabs" in m or m.get("ipcg")
+                    for m in extras["tensors"]
+                )
+                body_parts = [
+                    frames.encode_frame_prefix(frames.KIND_TENSOR, header)
+                ] + list(parts)
+            else:
+                body_parts = [
+                    frames.encode_frame_prefix(frames.KIND_PICKLE, header),
+                    parts[0],
+                ]
+        nbytes = sum(len(p) for p in body_parts)
+        return body_parts, extras, defer_ack, nbytes
 
     def _send_blocking(self, dest_party, data, up, down) -> bool:
         from rayfed_amd._private import tracing
@@ -194,42 +334,10 @@ class XferSenderService:
         try:
             if isinstance(data, ObjectRef):
                 data = data.result()  # producer error propagates to the future
-            from rayfed_amd.exceptions import FedRemoteError
-            from rayfed_amd.ops import shm_pool
-
-            header = {"job": self._job_name, "up": up, "down": down}
-            if isinstance(data, FedRemoteError):
-                from rayfed_amd._private import serialization
-
-                extras = {"tensors": []}
-                body_parts = [
-                    frames.encode_frame_prefix(frames.KIND_ERROR, header),
-                    serialization.dumps(data),
-                ]
-                defer_ack = False
-            else:
-                use_shm = shm_pool.shm_enabled() and self._same_host(dest_party)
-                extras, parts = tensor_codec.encode(data, self.gpu_plane, use_shm)
-                defer_ack = False
-                if extras["tensors"]:
-                    wire_header = {
-                        k: v for k, v in extras.items() if k != "_releases"
-                    }
-                    header.update(wire_header)
-                    defer_ack = any(
-                        "shm" in m or "ipc_slabs" in m or m.get("ipcg")
-                        for m in extras["tensors"]
-                    )
-                    body_parts = [
-                        frames.encode_frame_prefix(frames.KIND_TENSOR, header)
-                    ] + list(parts)
-                else:
-                    body_parts = [
-                        frames.encode_frame_prefix(frames.KIND_PICKLE, header),
-                        parts[0],
-                    ]
+            body_parts, extras, defer_ack, nbytes = self._encode_frame(
+                dest_party, data, up, down
+            )
             try:
-                nbytes = sum(len(p) for p in body_parts)
                 host, port = self._addresses[dest_party].rsplit(":", 1)
                 bulk = defer_ack or nbytes > (1 << 20)
                 code, result = self._send_with_retry(

@@ -46,7 +46,9 @@ def test_trace_file_written_and_parsable(tmp_path):
             names.add(e["name"])
             assert {"name", "cat", "ph", "ts", "pid", "tid"} <= set(e)
     assert "xsilo" in cats and "task" in cats
-    assert {"send", "recv", "xfer.send"} <= names
+    assert {"send", "recv"} <= names
+    # The transport emits either pooled spans or inline events per send.
+    assert names & {"xfer.send", "xfer.send_inline"}, names
 
 
 def test_tracing_disabled_by_default(tmp_path):
