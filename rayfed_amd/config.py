@@ -136,8 +136,13 @@ class GpuDataPlaneConfig:
     chunk_bytes: int = 64 << 20
     # Number of pinned staging buffers per direction (double/triple buffer).
     staging_buffers: int = 4
-    # Compute + verify CRC32 of every chunk on the GPU.
+    # Compute + verify a device checksum of every chunk on the GPU.
     verify_crc: bool = True
+    # Checksum algorithm for the device-IPC lane: "fnv64" (memory-rate
+    # parallel 64-bit hash — csrc hash64_kernel) or "crc32" (zlib-exact,
+    # LDS-lookup bound ~1.2 TB/s).  Host/socket/shm lanes always use CRC32
+    # (IO-bound there; zlib parity is useful on the wire).
+    device_checksum: str = "fnv64"
     # Receive tensors straight back onto the GPU of the consuming party.
     place_on_gpu: bool = True
     # Optional lossy wire compression for bf16 tensors: 'fp8e4m3' casts to

@@ -72,6 +72,32 @@ class GpuDataPlane:
         plane's device first (multi-GPU ranks pin one device per process)."""
         torch.cuda.set_device(self.device)
 
+    # -- device checksum dispatch --------------------------------------------
+    # The device-IPC lane defaults to hash64 (memory-rate FNV/murmur hash,
+    # csrc hash64_kernel); crc32 remains for 8-byte-misaligned views, for
+    # config device_checksum="crc32", and for every host/socket/shm lane.
+    # The algorithm rides the manifest so the receiver always verifies with
+    # the sender's choice.
+    def _ck_async(self, view: "torch.Tensor"):
+        """Launch the configured checksum kernel; returns (kind, out)."""
+        if (
+            self.config.device_checksum == "fnv64"
+            and (view.data_ptr() & 7) == 0
+        ):
+            return "fnv64", self._ext.hash64_async(view)
+        return "crc32", self._ext.crc32_async(view)
+
+    @staticmethod
+    def _ck_value(kind: str, out: "torch.Tensor") -> int:
+        if kind == "fnv64":
+            return int(out.item()) & 0xFFFFFFFFFFFFFFFF
+        return int(out[2].item()) & 0xFFFFFFFF
+
+    def _ck_verify_async(self, kind: str, view: "torch.Tensor"):
+        if kind == "fnv64":
+            return self._ext.hash64_async(view)
+        return self._ext.crc32_async(view)
+
     # -- buffer pools ---------------------------------------------------------
     def _get_buf(self, pool: List, nbytes: int, pin: bool) -> "torch.Tensor":
         with self._lock:
@@ -238,7 +264,7 @@ class GpuDataPlane:
                     slab[3][: hi - lo].copy_(flat[lo:hi])
                 if self.config.verify_crc:
                     with torch.cuda.stream(self._crc_stream):
-                        crc_outs.append(self._ext.crc32_async(flat[lo:hi]))
+                        crc_outs.append(self._ck_async(flat[lo:hi]))
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
@@ -246,14 +272,18 @@ class GpuDataPlane:
             with torch.cuda.stream(self._crc_stream):
                 crc_done = self._crc_stream.record_event()
             crc_done.synchronize()
-        crcs = (
-            [int(c[2].item()) & 0xFFFFFFFF for c in crc_outs]
-            if self.config.verify_crc
-            else None
-        )
+        crcs = cks = None
+        if self.config.verify_crc and crc_outs:
+            if wire_fp8:  # pack_fp8 computes CRC32 inline
+                crcs = [int(c[2].item()) & 0xFFFFFFFF for c in crc_outs]
+                cks = ["crc32"] * len(crc_outs)
+            else:
+                crcs = [self._ck_value(k, o) for k, o in crc_outs]
+                cks = [k for k, _ in crc_outs]
         man = {"ipc_slabs": [s[1] for s in slabs], "slab_bytes": S}
         if crcs is not None:
             man["ipc_crcs"] = crcs
+            man["cks"] = cks
         if wire_fp8:
             man["wire"] = "fp8e4m3"
 
@@ -278,7 +308,9 @@ class GpuDataPlane:
             if not t.is_contiguous():
                 t = t.contiguous()
             flats.append(t.view(-1).view(torch.uint8))
-        # Place tensors: bump offset, advance slab when one would straddle.
+        # Place tensors: bump offset (8-aligned so the receiver's checksum
+        # view can take the hash64 path), advance slab when one would
+        # straddle.
         placements = []
         slab_sizes: List[int] = []
         cur = 0
@@ -288,8 +320,8 @@ class GpuDataPlane:
                 slab_sizes.append(0)
                 cur = 0
             placements.append((len(slab_sizes) - 1, cur))
-            cur += n
-            slab_sizes[-1] = cur
+            cur = (cur + n + 7) & ~7
+            slab_sizes[-1] = min(cur, S)
         slabs = [self._ipc_get(sz) for sz in slab_sizes]
         produced = torch.cuda.current_stream(self.device).record_event()
         with torch.cuda.stream(self._copy_stream):
@@ -304,7 +336,7 @@ class GpuDataPlane:
                 slabs[si][3][off : off + n].copy_(f)
             if self.config.verify_crc:
                 with torch.cuda.stream(self._crc_stream):
-                    crc_outs.append(self._ext.crc32_async(f))
+                    crc_outs.append(self._ck_async(f))
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
@@ -317,7 +349,9 @@ class GpuDataPlane:
         for i, (si, off) in enumerate(placements):
             f = {"slab": si, "off": off}
             if self.config.verify_crc:
-                f["crc32"] = int(crc_outs[i][2].item()) & 0xFFFFFFFF
+                kind, out = crc_outs[i]
+                f["crc32"] = self._ck_value(kind, out)
+                f["ck"] = kind
             fields.append(f)
 
         def release():
@@ -336,6 +370,7 @@ class GpuDataPlane:
         src = self._ipc_src_view(handle, off + nbytes)[off : off + nbytes]
         out = torch.empty(shape, dtype=dtype, device=self.device)
         crc_expect = man.get("crc32")
+        ck = man.get("ck", "crc32")
         crc_out = None
         flat = out.view(-1).view(torch.uint8)
         with torch.cuda.stream(self._copy_stream):
@@ -343,15 +378,15 @@ class GpuDataPlane:
             done = self._copy_stream.record_event()
         if self.config.verify_crc and crc_expect is not None:
             with torch.cuda.stream(self._crc_stream):
-                crc_out = self._ext.crc32_async(src)
+                crc_out = self._ck_verify_async(ck, src)
                 crc_done = self._crc_stream.record_event()
         done.synchronize()
         if crc_out is not None:
             crc_done.synchronize()
-            got = int(crc_out[2].item()) & 0xFFFFFFFF
+            got = self._ck_value(ck, crc_out)
             if got != crc_expect:
                 raise ValueError(
-                    f"GPU tensor CRC mismatch (ipc group): expected "
+                    f"GPU tensor checksum mismatch (ipc group): expected "
                     f"{crc_expect:#x}, got {got:#x}"
                 )
         return out
@@ -373,27 +408,25 @@ class GpuDataPlane:
         S = man["slab_bytes"]
         handles = [bytes(h) for h in man["ipc_slabs"]]
         crcs = man.get("ipc_crcs")
+        cks = man.get("cks") or ["crc32"] * len(handles)
         wire_fp8 = man.get("wire") == "fp8e4m3"
         out = torch.empty(shape, dtype=dtype, device=self.device)
         crc_outs = []
-        # Copy ∥ CRC on separate streams (see pack_to_ipc): the D2D copy runs
-        # at memory rate while the slower CRC pass verifies the same source.
+        # Copy ∥ checksum on separate streams (see pack_to_ipc): the D2D copy
+        # runs at memory rate while the verify pass reads the same source.
         for i, h in enumerate(handles):
             lo, hi = i * S, min((i + 1) * S, nbytes)
             src = self._ipc_src_view(h, hi - lo)[: hi - lo]
             if wire_fp8:
                 with torch.cuda.stream(self._copy_stream):
                     self._ext.unpack_fp8_async(src, out.view(-1)[lo:hi])
-                if self.config.verify_crc and crcs is not None:
-                    with torch.cuda.stream(self._crc_stream):
-                        crc_outs.append(self._ext.crc32_async(src))
             else:
                 flat = out.view(-1).view(torch.uint8)
                 with torch.cuda.stream(self._copy_stream):
                     flat[lo:hi].copy_(src)
-                if self.config.verify_crc and crcs is not None:
-                    with torch.cuda.stream(self._crc_stream):
-                        crc_outs.append(self._ext.crc32_async(src))
+            if self.config.verify_crc and crcs is not None:
+                with torch.cuda.stream(self._crc_stream):
+                    crc_outs.append(self._ck_verify_async(cks[i], src))
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
@@ -402,11 +435,11 @@ class GpuDataPlane:
                 crc_done = self._crc_stream.record_event()
             crc_done.synchronize()
         for i, c in enumerate(crc_outs):
-            got = int(c[2].item()) & 0xFFFFFFFF
+            got = self._ck_value(cks[i], c)
             if got != crcs[i]:
                 raise ValueError(
-                    f"GPU tensor CRC mismatch (ipc lane, slab {i}): expected "
-                    f"{crcs[i]:#x}, got {got:#x}"
+                    f"GPU tensor checksum mismatch (ipc lane, slab {i}): "
+                    f"expected {crcs[i]:#x}, got {got:#x}"
                 )
         return out
 
