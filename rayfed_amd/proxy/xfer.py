@@ -220,14 +220,18 @@ class XferSenderService:
                     except RuntimeError:
                         # e.g. peer not up yet: the pooled path carries the
                         # retry/backoff budget (async-startup semantics).
-                        tensor_codec.release_parts(extras)
-                        state.clear()
+                        state.pop("handle", None)
                         state["fut"] = self._pool.submit(
-                            self._send_blocking, dest_party, d, up, down
+                            self._send_parts, dest_party, body_parts,
+                            extras, False, nbytes, up, down,
                         )
                 else:
+                    # Big / DEFER_ACK frame: ship the ALREADY-ENCODED parts
+                    # on the pool — re-encoding would pack (and leak) a
+                    # second set of staging slabs.
                     state["fut"] = self._pool.submit(
-                        self._send_blocking, dest_party, d, up, down
+                        self._send_parts, dest_party, body_parts, extras,
+                        defer_ack, nbytes, up, down,
                     )
             except BaseException as e:  # noqa: BLE001
                 state["exc"] = e
@@ -328,15 +332,28 @@ class XferSenderService:
         return self._send_blocking_inner(dest_party, data, up, down)
 
     def _send_blocking_inner(self, dest_party, data, up, down) -> bool:
-        t0 = time.perf_counter()
-        err = True
-        nbytes = 0
+        if isinstance(data, ObjectRef):
+            data = data.result()  # producer error propagates to the future
         try:
-            if isinstance(data, ObjectRef):
-                data = data.result()  # producer error propagates to the future
             body_parts, extras, defer_ack, nbytes = self._encode_frame(
                 dest_party, data, up, down
             )
+        except BaseException:
+            with self._stats_lock:
+                edge = self._edges.setdefault(dest_party, self._edge_cls())
+                edge.record(0, 0.0, True)
+            raise
+        return self._send_parts(
+            dest_party, body_parts, extras, defer_ack, nbytes, up, down
+        )
+
+    def _send_parts(self, dest_party, body_parts, extras, defer_ack, nbytes,
+                    up, down) -> bool:
+        """Ship pre-encoded frame parts with retry; releases staging after
+        the ack and records edge stats."""
+        t0 = time.perf_counter()
+        err = True
+        try:
             try:
                 host, port = self._addresses[dest_party].rsplit(":", 1)
                 bulk = defer_ack or nbytes > (1 << 20)
