@@ -53,6 +53,13 @@ namespace py = pybind11;
 namespace {
 
 constexpr uint8_t kFlagDeferAck = 1;
+// STRIPE sub-frame: one shard of a large body, sent over its own
+// connection in parallel with its siblings.  The preamble gains
+// u32 idx ‖ u32 cnt ‖ u64 off ‖ u64 body_total after the seq ids; the
+// receiver's connection threads read each stripe's payload DIRECTLY into
+// the shared assembly buffer at `off` (no reassembly copy) and the last
+// stripe posts the completed body to the mailbox.
+constexpr uint8_t kFlagStripe = 2;
 // Reserved seq id of the init-time readiness barrier (constants.PING_SEQ_ID):
 // acked without parking so repeated pings never leak mailbox slots.
 constexpr const char* kPingSeqId = "ping";
@@ -396,6 +403,29 @@ class XferServer {
     std::atomic<bool> alive{true};
   };
 
+  // In-flight striped body, filled by several connection threads at once.
+  // `got` counts bytes landed; the stripe that completes the byte count
+  // moves the body to the mailbox.  (A retried frame after a partial
+  // failure recreates the slot only if its geometry changed; a stale mix
+  // is caught by the payload checksum downstream.)
+  struct Assembly {
+    std::string data;
+    std::atomic<uint64_t> got{0};
+    uint64_t total;
+    uint32_t cnt;
+    Assembly(uint64_t t, uint32_t c) : total(t), cnt(c) { data.resize(t); }
+  };
+
+  static bool drain(Stream& st, uint64_t n) {
+    char scratch[65536];
+    while (n) {
+      size_t k = (size_t)std::min<uint64_t>(n, sizeof(scratch));
+      if (!st.read_all(scratch, k)) return false;
+      n -= k;
+    }
+    return true;
+  }
+
   static void write_ack(SrvConn& c, uint64_t req_id, uint16_t code,
                         const std::string& result) {
     char ack[14];
@@ -446,23 +476,77 @@ class XferServer {
       memcpy(&total, head, 8);
       memcpy(&req_id, head + 8, 8);
       if (total < 12 || total > (64ull << 30)) break;  // sane bounds
-      buf.resize(total - 8);  // everything after req_id
-      if (!st.read_all(buf.data(), buf.size())) break;
-      uint8_t flags = (uint8_t)buf[0];
-      uint8_t job_len = (uint8_t)buf[1];
-      uint8_t up_len = (uint8_t)buf[2];
-      uint8_t down_len = (uint8_t)buf[3];
-      size_t off = 4;
-      if (off + job_len + up_len + down_len > buf.size()) break;
-      std::string job(buf.data() + off, job_len);
-      off += job_len;
-      std::string up(buf.data() + off, up_len);
-      off += up_len;
-      std::string down(buf.data() + off, down_len);
-      off += down_len;
+      char meta[4];
+      if (!st.read_all(meta, 4)) break;
+      const uint8_t flags = (uint8_t)meta[0];
+      const uint8_t job_len = (uint8_t)meta[1];
+      const uint8_t up_len = (uint8_t)meta[2];
+      const uint8_t down_len = (uint8_t)meta[3];
+      const size_t names_len = (size_t)job_len + up_len + down_len;
+      const size_t fixed =
+          8 + 4 + names_len + ((flags & kFlagStripe) ? 24u : 0u);
+      if (total < fixed) break;
+      char names[768];
+      if (!st.read_all(names, names_len)) break;
+      std::string job(names, job_len);
+      std::string up(names + job_len, up_len);
+      std::string down(names + job_len + up_len, down_len);
+      const uint64_t payload_len = total - fixed;
 
       uint16_t code = 200;
       std::string result = "OK";
+      if (flags & kFlagStripe) {
+        char smeta[24];
+        if (!st.read_all(smeta, 24)) break;
+        uint32_t idx, cnt;
+        uint64_t s_off, body_total;
+        memcpy(&idx, smeta, 4);
+        memcpy(&cnt, smeta + 4, 4);
+        memcpy(&s_off, smeta + 8, 8);
+        memcpy(&body_total, smeta + 16, 8);
+        if (cnt == 0 || idx >= cnt || body_total > (64ull << 30) ||
+            s_off + payload_len > body_total)
+          break;
+        if (job != job_) {
+          if (!drain(st, payload_len)) break;
+          code = 417;
+          result = "JobName mis-match: expected " + job_ + ", got " + job;
+        } else {
+          std::string key = up + '\x00' + down;
+          std::shared_ptr<Assembly> asmb;
+          {
+            std::lock_guard<std::mutex> lk(asm_mu_);
+            auto& slot = asm_[key];
+            if (!slot || slot->total != body_total || slot->cnt != cnt)
+              slot = std::make_shared<Assembly>(body_total, cnt);
+            asmb = slot;
+          }
+          // Payload streams DIRECTLY into the assembly buffer at its
+          // offset — stripes of one frame write disjoint ranges from
+          // their own connection threads, no reassembly copy.
+          if (!st.read_all(asmb->data.data() + s_off, payload_len)) break;
+          if (asmb->got.fetch_add(payload_len) + payload_len == body_total) {
+            {
+              std::lock_guard<std::mutex> lk(asm_mu_);
+              asm_.erase(key);
+            }
+            recv_count_.fetch_add(1);
+            std::lock_guard<std::mutex> lk(mail_mu_);
+            mail_[key] = std::move(asmb->data);
+            mail_cv_.notify_all();
+          }
+        }
+        try {
+          write_ack(*conn, req_id, code, result);
+        } catch (...) {
+          break;
+        }
+        continue;
+      }
+      buf.resize(payload_len);
+      if (!st.read_all(buf.data(), payload_len)) break;
+      size_t off = 0;
+
       if (job != job_) {
         code = 417;
         result = "JobName mis-match: expected " + job_ + ", got " + job;
@@ -525,6 +609,8 @@ class XferServer {
   std::mutex mail_mu_;
   std::condition_variable mail_cv_;
   std::map<std::string, std::string> mail_;
+  std::mutex asm_mu_;
+  std::unordered_map<std::string, std::shared_ptr<Assembly>> asm_;
   std::atomic<uint64_t> recv_count_{0};
   std::mutex tok_mu_;
   std::unordered_map<uint64_t, std::pair<std::shared_ptr<SrvConn>, uint64_t>>
@@ -550,11 +636,170 @@ class XferClient {
     if (ssl_ctx_) SSL_CTX_free(ssl_ctx_);
   }
 
+  // Slice a scatter-gather view list down to the byte range [lo, hi).
+  static std::vector<std::pair<const char*, size_t>> slice_views(
+      const std::vector<std::pair<const char*, size_t>>& views, uint64_t lo,
+      uint64_t hi) {
+    std::vector<std::pair<const char*, size_t>> out;
+    uint64_t pos = 0;
+    for (const auto& v : views) {
+      uint64_t vlo = pos, vhi = pos + v.second;
+      pos = vhi;
+      if (vhi <= lo || vlo >= hi) continue;
+      uint64_t s = std::max(vlo, lo), e = std::min(vhi, hi);
+      out.emplace_back(v.first + (s - vlo), (size_t)(e - s));
+    }
+    return out;
+  }
+
+  // Striped send: split the body across `stripes` parallel connections —
+  // a single TCP stream tops out ~2 GB/s on this path; N streams with the
+  // receiver assembling in place scale it.  Blocks until every stripe is
+  // acked; any stripe failure fails the send (caller retries whole-frame).
+  int send_striped(const std::string& host, int port, const std::string& up,
+                   const std::string& down,
+                   const std::vector<std::pair<const char*, size_t>>& views,
+                   uint64_t body_len, int stripes, double timeout_s,
+                   std::string* result_out) {
+    std::string names;
+    names.push_back((char)(kFlagStripe));
+    names.push_back((char)job_.size());
+    names.push_back((char)up.size());
+    names.push_back((char)down.size());
+    names += job_;
+    names += up;
+    names += down;
+    const uint64_t chunk = (body_len + stripes - 1) / stripes;
+    std::vector<int> codes((size_t)stripes, 0);
+    std::vector<std::string> results((size_t)stripes);
+    std::vector<std::string> errs((size_t)stripes);
+    std::vector<std::thread> threads;
+    for (int i = 0; i < stripes; ++i) {
+      threads.emplace_back([&, i] {
+        try {
+          const uint64_t lo = (uint64_t)i * chunk;
+          const uint64_t hi = std::min(body_len, lo + chunk);
+          auto sub = slice_views(views, lo, hi);
+          char smeta[24];
+          uint32_t idx = (uint32_t)i, cnt = (uint32_t)stripes;
+          memcpy(smeta, &idx, 4);
+          memcpy(smeta + 4, &cnt, 4);
+          memcpy(smeta + 8, &lo, 8);
+          memcpy(smeta + 16, &body_len, 8);
+          uint64_t total = 8 + names.size() + 24 + (hi - lo);
+          if (ssl_ctx_) {
+            codes[i] = stripe_tls(host, port, names, smeta, sub, total,
+                                  timeout_s, &results[i]);
+            return;
+          }
+          std::shared_ptr<Conn> conn_sp = get_conn(host, port, i + 1);
+          Conn& conn = *conn_sp;
+          std::shared_ptr<Pending> pending = std::make_shared<Pending>();
+          uint64_t req_id;
+          {
+            std::lock_guard<std::mutex> lk(conn.write_mu);
+            req_id = conn.next_id++;
+            {
+              std::lock_guard<std::mutex> lk2(conn.pend_mu);
+              conn.pending[req_id] = pending;
+            }
+            char head[16];
+            memcpy(head, &total, 8);
+            memcpy(head + 8, &req_id, 8);
+            std::vector<iovec> iov;
+            iov.push_back({head, 16});
+            iov.push_back({(void*)names.data(), names.size()});
+            iov.push_back({smeta, 24});
+            for (auto& v : sub) iov.push_back({(void*)v.first, v.second});
+            try {
+              writev_all(conn.fd, std::move(iov));
+            } catch (...) {
+              conn.alive = false;
+              throw;
+            }
+          }
+          std::unique_lock<std::mutex> lk(pending->mu);
+          auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::duration<double>(timeout_s);
+          while (!pending->done) {
+            if (pending->cv.wait_until(lk, deadline) ==
+                std::cv_status::timeout)
+              throw std::runtime_error("stripe ack timeout");
+          }
+          if (pending->broken)
+            throw std::runtime_error("connection broken awaiting stripe ack");
+          codes[i] = pending->code;
+          results[i] = pending->result;
+        } catch (const std::exception& e) {
+          errs[i] = e.what();
+        }
+      });
+    }
+    for (auto& t : threads) t.join();
+    for (int i = 0; i < stripes; ++i)
+      if (!errs[i].empty())
+        throw std::runtime_error("stripe " + std::to_string(i) +
+                                 " failed: " + errs[i]);
+    for (int i = 0; i < stripes; ++i) {
+      if (codes[i] != 200) {
+        if (result_out) *result_out = results[i];
+        return codes[i];
+      }
+    }
+    if (result_out) *result_out = "OK";
+    return 200;
+  }
+
+  int stripe_tls(const std::string& host, int port, const std::string& names,
+                 const char* smeta,
+                 const std::vector<std::pair<const char*, size_t>>& sub,
+                 uint64_t total, double timeout_s, std::string* result_out) {
+    std::string key = host + ":" + std::to_string(port);
+    std::unique_ptr<Stream> st = acquire_tls_conn(key, host, port);
+    uint64_t req_id = tls_req_id_.fetch_add(1);
+    try {
+      char head[16];
+      memcpy(head, &total, 8);
+      memcpy(head + 8, &req_id, 8);
+      std::vector<iovec> iov;
+      iov.push_back({head, 16});
+      iov.push_back({(void*)names.data(), names.size()});
+      iov.push_back({(void*)smeta, 24});
+      for (auto& v : sub) iov.push_back({(void*)v.first, v.second});
+      st->write_iov(std::move(iov));
+      char ahead[14];
+      if (!st->read_all(ahead, 14))
+        throw std::runtime_error("connection broken awaiting stripe ack");
+      uint32_t len;
+      uint64_t rid;
+      uint16_t code;
+      memcpy(&len, ahead, 4);
+      memcpy(&rid, ahead + 4, 8);
+      memcpy(&code, ahead + 12, 2);
+      std::string result;
+      if (len > 10) {
+        result.resize(len - 10);
+        if (!st->read_all(result.data(), result.size()))
+          throw std::runtime_error("connection broken reading stripe ack");
+      }
+      if (rid != req_id)
+        throw std::runtime_error("ack id mismatch on TLS stripe");
+      release_tls_conn(key, std::move(st));
+      if (result_out) *result_out = result;
+      return code;
+    } catch (...) {
+      st->close_free();
+      throw;
+    }
+  }
+
   // Blocking send with ack round trip; GIL released around I/O.
   // parts: list of buffer-likes written scatter-gather (no join copy).
+  // stripes > 1 splits the body across that many parallel connections.
   int send(const std::string& host, int port, const std::string& up,
            const std::string& down, std::vector<py::buffer> parts,
-           bool defer_ack, double timeout_s, std::string* result_out) {
+           bool defer_ack, double timeout_s, std::string* result_out,
+           int stripes = 1) {
     // Collect buffer pointers under the GIL.
     std::vector<std::pair<const char*, size_t>> views;
     views.reserve(parts.size());
@@ -567,6 +812,12 @@ class XferClient {
       views.emplace_back((const char*)info.ptr,
                          (size_t)(info.size * info.itemsize));
       body_len += views.back().second;
+    }
+
+    if (stripes > 1 && !defer_ack && body_len > 0) {
+      py::gil_scoped_release release;
+      return send_striped(host, port, up, down, views, body_len, stripes,
+                          timeout_s, result_out);
     }
 
     uint8_t flags = defer_ack ? kFlagDeferAck : 0;
@@ -866,8 +1117,10 @@ class XferClient {
     std::atomic<bool> alive{true};
   };
 
-  std::shared_ptr<Conn> get_conn(const std::string& host, int port) {
+  std::shared_ptr<Conn> get_conn(const std::string& host, int port,
+                                 int slot = 0) {
     std::string key = host + ":" + std::to_string(port);
+    if (slot) key += "#" + std::to_string(slot);
     // Serialize (re)connection per client: the lock is held through
     // connect() so concurrent first-sends share one socket, and dead conns
     // are REPLACED, never destroyed, while a sender still references them
@@ -898,13 +1151,13 @@ class XferClient {
     conn->fd = fd;
     std::shared_ptr<Conn> sp = conn;
     conn->reader = std::thread([this, sp] { reader_loop(sp.get()); });
-    auto& slot = conns_[key];
-    if (slot) {  // replace a dead conn; readers/senders keep their refs
-      ::shutdown(slot->fd, SHUT_RDWR);
-      if (slot->reader.joinable()) slot->reader.detach();
-      dead_.push_back(slot);
+    auto& entry = conns_[key];
+    if (entry) {  // replace a dead conn; readers/senders keep their refs
+      ::shutdown(entry->fd, SHUT_RDWR);
+      if (entry->reader.joinable()) entry->reader.detach();
+      dead_.push_back(entry);
     }
-    slot = conn;
+    entry = conn;
     return conn;
   }
 
@@ -996,16 +1249,16 @@ PYBIND11_MODULE(_xfer, m) {
           [](XferClient& c, const std::string& host, int port,
              const std::string& up, const std::string& down,
              std::vector<py::buffer> parts, bool defer_ack,
-             double timeout_s) {
+             double timeout_s, int stripes) {
             std::string result;
             int code =
                 c.send(host, port, up, down, std::move(parts), defer_ack,
-                       timeout_s, &result);
+                       timeout_s, &result, stripes);
             return py::make_tuple(code, result);
           },
           py::arg("host"), py::arg("port"), py::arg("up"), py::arg("down"),
           py::arg("parts"), py::arg("defer_ack") = false,
-          py::arg("timeout_s") = 60.0)
+          py::arg("timeout_s") = 60.0, py::arg("stripes") = 1)
       .def("send_async", &XferClient::send_async, py::arg("host"),
            py::arg("port"), py::arg("up"), py::arg("down"), py::arg("parts"),
            py::arg("defer_ack") = false)

@@ -10,7 +10,8 @@ generic-handler API with identity (de)serializers:
 - the (potentially multi-GiB) tensor payload is never copied into a protobuf
   message — the frame is ``header || payload`` and the payload slice is a
   zero-copy ``memoryview`` on the receive side;
-- a streaming variant carries chunked tensor payloads for D2H/network overlap.
+- large bodies stream as parallel STRIPE sub-frames at the transport layer
+  (csrc/xfer_core.cpp) — no frame-level streaming variant is needed.
 
 Frame layout (little-endian)::
 
@@ -43,7 +44,6 @@ _PREFIX = struct.Struct("<4sBBHI")
 
 SERVICE_NAME = "rayfedamd.GrpcService"
 SEND_DATA_METHOD = f"/{SERVICE_NAME}/SendData"
-SEND_STREAM_METHOD = f"/{SERVICE_NAME}/SendStream"
 
 
 def encode_frame(

@@ -134,13 +134,14 @@ class _Connection:
                 fut.set_exception(ConnectionError(str(exc)))
         self.pending.clear()
 
-    async def request(self, prefix: bytes, parts, timeout: float) -> dict:
+    async def request(self, prefix: bytes, parts, timeout: float,
+                      chunk: int = 8 << 20) -> dict:
         req_id = next(self.ids)
         fut = asyncio.get_running_loop().create_future()
         self.pending[req_id] = fut
         total = len(prefix) + sum(len(p) for p in parts)
         head = (total + 8).to_bytes(8, "little") + req_id.to_bytes(8, "little")
-        _CHUNK = 8 << 20  # bound transport buffering for multi-GiB parts
+        _CHUNK = chunk  # bound transport buffering for multi-GiB parts
         async with self.write_lock:  # one frame's bytes stay contiguous
             if total <= 65536:
                 # Small request: one write, one TCP segment.
@@ -192,6 +193,10 @@ class TcpSenderProxy(base_proxy.SenderProxy):
             if proxy_config and proxy_config.timeout_in_ms
             else 60.0
         )
+        # messages_max_size_in_bytes chunks this transport's writes (the
+        # reference uses it as the gRPC message cap, grpc_options.py:28-29).
+        cap = proxy_config.messages_max_size_in_bytes if proxy_config else None
+        self._write_chunk = min(8 << 20, cap) if cap else 8 << 20
         self._ssl = _client_ssl_context(tls_config) if tls_config else None
         # Verify the certificate against the destination host unless the user
         # pins a name explicitly (same semantics as the reference's
@@ -264,7 +269,10 @@ class TcpSenderProxy(base_proxy.SenderProxy):
                 remaining = deadline - time.monotonic()
                 if remaining <= 0:
                     raise TimeoutError(f"send to {dest_party} deadline exceeded")
-                resp = await conn.request(req.prefix, req.parts, timeout=remaining)
+                resp = await conn.request(
+                    req.prefix, req.parts, timeout=remaining,
+                    chunk=self._write_chunk,
+                )
                 code = resp.get("code", 500)
                 if 400 <= code < 500:
                     raise RuntimeError(

@@ -146,6 +146,15 @@ class XferSenderService:
             if proxy_config and proxy_config.timeout_in_ms
             else 60.0
         )
+        # Frames above one stripe's worth split across parallel connections
+        # (cross-host lane: one TCP stream tops out ~2 GB/s).  The stripe
+        # size honors messages_max_size_in_bytes when set — the message-cap
+        # knob chunks on this transport like it caps gRPC messages.
+        self._stripe_bytes = (
+            proxy_config.messages_max_size_in_bytes
+            if proxy_config and proxy_config.messages_max_size_in_bytes
+            else (256 << 20)
+        )
         self.gpu_plane = None
         self._stats_lock = threading.Lock()
         self.send_op_count = 0
@@ -357,9 +366,13 @@ class XferSenderService:
             try:
                 host, port = self._addresses[dest_party].rsplit(":", 1)
                 bulk = defer_ack or nbytes > (1 << 20)
+                stripes = 1
+                if not defer_ack and nbytes >= 2 * self._stripe_bytes:
+                    stripes = min(8, -(-nbytes // self._stripe_bytes))
                 code, result = self._send_with_retry(
                     self._client_bulk if bulk else self._client_ctl,
                     host, int(port), up, down, body_parts, defer_ack,
+                    stripes=stripes,
                 )
             finally:
                 tensor_codec.release_parts(extras)
@@ -379,7 +392,8 @@ class XferSenderService:
                 edge = self._edges.setdefault(dest_party, self._edge_cls())
                 edge.record(nbytes, secs, err)
 
-    def _send_with_retry(self, client, host, port, up, down, parts, defer_ack):
+    def _send_with_retry(self, client, host, port, up, down, parts, defer_ack,
+                         stripes: int = 1):
         deadline = time.monotonic() + self._timeout_s
         backoff = self._retry.initial_backoff
         attempt = 0
@@ -388,7 +402,8 @@ class XferSenderService:
             try:
                 remaining = max(0.001, deadline - time.monotonic())
                 return client.send(
-                    host, port, up, down, parts, defer_ack, remaining
+                    host, port, up, down, parts, defer_ack, remaining,
+                    stripes,
                 )
             except RuntimeError as e:
                 now = time.monotonic()
