@@ -668,6 +668,34 @@ class GpuDataPlane:
             crc = int(crc_out[2].item()) & 0xFFFFFFFF
         return seg, crc, lambda: pool.release(seg)
 
+    # -- streamed H2D (chunked socket lane) ----------------------------------
+    def h2d_copy(self, dst_flat_u8: "torch.Tensor", dst_off: int, data):
+        """Stage ``data`` through a pooled pinned buffer and H2D it into
+        ``dst_flat_u8[dst_off:]`` asynchronously on the copy stream.  The
+        CPU memcpy of the NEXT chunk overlaps this chunk's DMA.  Returns
+        (event, pinned) for :meth:`finish_h2d`."""
+        self._bind_device()
+        n = len(data)
+        pinned = self._get_buf(self._pinned, n, pin=True)
+        pinned[:n].numpy()[:] = memoryview(data).cast("B")
+        with torch.cuda.stream(self._copy_stream):
+            dst_flat_u8[dst_off : dst_off + n].copy_(
+                pinned[:n], non_blocking=True
+            )
+            ev = torch.cuda.Event()
+            ev.record(self._copy_stream)
+        return ev, pinned
+
+    def finish_h2d(self, pend) -> None:
+        for ev, pinned in pend:
+            ev.synchronize()
+            self._put_buf(self._pinned, pinned)
+
+    def device_crc32(self, flat_u8: "torch.Tensor") -> int:
+        out = self._ext.crc32_async(flat_u8)
+        torch.cuda.current_stream(self.device).synchronize()
+        return int(out[2].item()) & 0xFFFFFFFF
+
     # -- recv path ------------------------------------------------------------
     def unpack_from_host(
         self,

@@ -293,6 +293,90 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
     return extras, parts
 
 
+def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
+                    chunks_iter, gpu_plane=None, allowed_list=None):
+    """Decode a chunk-streamed payload (KIND_CHUNKED): ``chunks_iter``
+    yields ``(index, bytes)`` in ARRIVAL order; chunk i covers payload
+    bytes [i*chunk_bytes, ...).  GPU-destined payload-route tensors H2D
+    each chunk as it lands — the copy-in overlaps the network — and verify
+    their checksum on device afterwards.  Any layout this fast path does
+    not handle (shm/ipc manifests, fp8 wire, CPU destination) falls back
+    to assemble-then-decode.
+    """
+    mans = header["tensors"]
+    skel_len = header["skel"]
+    streamable = (
+        torch is not None
+        and gpu_plane is not None
+        and gpu_plane.config.place_on_gpu
+        and all(
+            "shm" not in m and "ipc_slabs" not in m and not m.get("ipcg")
+            and not m.get("wire") and m["device"] == "cuda"
+            for m in mans
+        )
+        and mans
+    )
+    if not streamable:
+        buf = bytearray(total_len)
+        for i, data in chunks_iter:
+            lo = i * chunk_bytes
+            buf[lo : lo + len(data)] = data
+        return decode(header, memoryview(buf), gpu_plane, allowed_list)
+
+    # Payload spans per tensor: skeleton first, then raw bytes in order.
+    spans = []
+    off = skel_len
+    outs = []
+    for m in mans:
+        n = m["nbytes"]
+        spans.append((off, off + n))
+        off += n
+        outs.append(
+            torch.empty(
+                m["shape"], dtype=_STR_TO_DTYPE[m["dtype"]],
+                device=gpu_plane.device,
+            )
+        )
+    skeleton = bytearray(skel_len)
+    pend = []  # (event, pinned) — released after the final sync
+    for i, data in chunks_iter:
+        lo = i * chunk_bytes
+        hi = lo + len(data)
+        mv = memoryview(data)
+        if lo < skel_len:
+            take = min(hi, skel_len) - lo
+            skeleton[lo : lo + take] = mv[:take]
+        for (slo, shi), out in zip(spans, outs):
+            if hi <= slo or lo >= shi:
+                continue
+            s = max(lo, slo)
+            e = min(hi, shi)
+            flat = out.view(-1).view(torch.uint8)
+            pend.append(
+                gpu_plane.h2d_copy(flat, s - slo, mv[s - lo : e - lo])
+            )
+    gpu_plane.finish_h2d(pend)
+    # Device-side verify against the sender's wire checksum.
+    if gpu_plane.config.verify_crc:
+        for m, out in zip(mans, outs):
+            if m.get("crc32") is None:
+                continue
+            flat = out.view(-1).view(torch.uint8)
+            got = gpu_plane.device_crc32(flat)
+            if got != m["crc32"]:
+                raise ValueError(
+                    f"tensor CRC mismatch (streamed): expected "
+                    f"{m['crc32']:#x}, got {got:#x}"
+                )
+    _decode_ctx.tensors = outs
+    try:
+        from rayfed_amd._private import serialization
+
+        return serialization.loads(bytes(skeleton), allowed_list)
+    finally:
+        _decode_ctx.tensors = None
+
+
 def release_parts(extras: Dict) -> None:
     """Return pooled staging buffers referenced by encode() output."""
     for rel in extras.pop("_releases", []):

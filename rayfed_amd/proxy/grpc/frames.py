@@ -39,6 +39,11 @@ VERSION = 1
 KIND_PICKLE = 0
 KIND_TENSOR = 1
 KIND_ERROR = 2
+# Chunk-streamed tensor frame (C++ transport only): the main frame's payload
+# is the INNER tensor frame's prefix; the payload bytes travel as xk sidecar
+# frames under derived seq ids, consumed progressively (H2D overlaps network
+# arrival — tensor_codec.decode_streamed).
+KIND_CHUNKED = 3
 
 _PREFIX = struct.Struct("<4sBBHI")
 

@@ -135,6 +135,10 @@ class XferSenderService:
             max_workers=workers, thread_name_prefix="xfer-send"
         )
         _prestart_pool(self._pool, workers)
+        # Sidecar chunk frames of chunk-streamed sends (see _send_chunked).
+        self._chunk_pool = ThreadPoolExecutor(
+            max_workers=8, thread_name_prefix="xfer-chunk"
+        )
         self._retry = _Retry(
             getattr(proxy_config, "grpc_retry_policy", None) if proxy_config else None,
             proxy_max_restarts=(
@@ -356,6 +360,20 @@ class XferSenderService:
             dest_party, body_parts, extras, defer_ack, nbytes, up, down
         )
 
+    @staticmethod
+    def _slice_parts(parts, lo, hi):
+        """Byte range [lo, hi) across a list of buffer parts (zero-copy)."""
+        out = []
+        pos = 0
+        for p in parts:
+            mv = memoryview(p)
+            plo, phi = pos, pos + len(mv)
+            pos = phi
+            if phi <= lo or plo >= hi:
+                continue
+            out.append(mv[max(plo, lo) - plo : min(phi, hi) - plo])
+        return out
+
     def _send_parts(self, dest_party, body_parts, extras, defer_ack, nbytes,
                     up, down) -> bool:
         """Ship pre-encoded frame parts with retry; releases staging after
@@ -365,15 +383,32 @@ class XferSenderService:
         try:
             try:
                 host, port = self._addresses[dest_party].rsplit(":", 1)
-                bulk = defer_ack or nbytes > (1 << 20)
-                stripes = 1
-                if not defer_ack and nbytes >= 2 * self._stripe_bytes:
-                    stripes = min(8, -(-nbytes // self._stripe_bytes))
-                code, result = self._send_with_retry(
-                    self._client_bulk if bulk else self._client_ctl,
-                    host, int(port), up, down, body_parts, defer_ack,
-                    stripes=stripes,
-                )
+                prefix = bytes(body_parts[0][:6]) if body_parts else b""
+                payload_len = nbytes - (len(body_parts[0]) if body_parts else 0)
+                if (
+                    not defer_ack
+                    and not self._tls
+                    and len(prefix) == 6
+                    and prefix[:4] == frames.MAGIC
+                    and prefix[5] == frames.KIND_TENSOR
+                    and payload_len >= 2 * self._stripe_bytes
+                ):
+                    # Chunk-streamed tensor frame: the receiver H2Ds each
+                    # chunk as it lands, overlapping consume with arrival
+                    # (a whole-frame send serializes network then consume).
+                    code, result = self._send_chunked(
+                        host, int(port), up, down, body_parts, payload_len
+                    )
+                else:
+                    bulk = defer_ack or nbytes > (1 << 20)
+                    stripes = 1
+                    if not defer_ack and nbytes >= 2 * self._stripe_bytes:
+                        stripes = min(8, -(-nbytes // self._stripe_bytes))
+                    code, result = self._send_with_retry(
+                        self._client_bulk if bulk else self._client_ctl,
+                        host, int(port), up, down, body_parts, defer_ack,
+                        stripes=stripes,
+                    )
             finally:
                 tensor_codec.release_parts(extras)
             if 400 <= code < 500:
@@ -391,6 +426,42 @@ class XferSenderService:
             with self._stats_lock:
                 edge = self._edges.setdefault(dest_party, self._edge_cls())
                 edge.record(nbytes, secs, err)
+
+    def _send_chunked(self, host, port, up, down, body_parts, payload_len):
+        """Split a tensor frame's payload into sidecar chunk frames sent in
+        parallel (each striped over the stripe connections), plus a small
+        KIND_CHUNKED main frame carrying the inner frame's prefix."""
+        C = self._stripe_bytes
+        K = min(32, -(-payload_len // C))
+        C = -(-payload_len // K)  # rebalance so every chunk is ~equal
+        payload_parts = list(body_parts[1:])
+        futs = []
+        for i in range(K):
+            sub = self._slice_parts(payload_parts, i * C, min((i + 1) * C,
+                                                              payload_len))
+            # Dedicated pool: _send_parts itself runs on self._pool, and
+            # waiting there for chunk futures queued to the same pool could
+            # starve under concurrent big sends.
+            futs.append(self._chunk_pool.submit(
+                self._send_with_retry, self._client_bulk, host, port,
+                f"{up}\x01x{i}", down, sub, False, 4,
+            ))
+        meta = {
+            "job": self._job_name, "up": up, "down": down,
+            "xk": K, "xc": C, "xlen": payload_len,
+        }
+        main = [
+            frames.encode_frame_prefix(frames.KIND_CHUNKED, meta),
+            body_parts[0],
+        ]
+        code, result = self._send_with_retry(
+            self._client_ctl, host, port, up, down, main, False
+        )
+        for f in futs:
+            c, r = f.result()
+            if c != 200 and code == 200:
+                code, result = c, r
+        return code, result
 
     def _send_with_retry(self, client, host, port, up, down, parts, defer_ack,
                          stripes: int = 1):
@@ -425,6 +496,7 @@ class XferSenderService:
 
     def stop(self):
         self._pool.shutdown(wait=False, cancel_futures=True)
+        self._chunk_pool.shutdown(wait=False, cancel_futures=True)
         self._client_ctl.close_all()
         self._client_bulk.close_all()
 
@@ -526,7 +598,50 @@ class XferReceiverService:
             return tensor_codec.decode(
                 header, memoryview(payload), self.gpu_plane, self._allowed_list
             )
+        if kind == frames.KIND_CHUNKED:
+            _ik, inner_header, _ = frames.decode_frame(payload)
+            return tensor_codec.decode_streamed(
+                inner_header, header["xlen"], header["xc"],
+                self._iter_chunks(up, down, header["xk"]),
+                self.gpu_plane, self._allowed_list,
+            )
         return serialization.loads(payload, self._allowed_list)
+
+    def _iter_chunks(self, up: str, down: str, k: int):
+        """Yield sidecar chunk frames of a chunk-streamed send in ARRIVAL
+        order (consume overlaps the network).  If the sender aborts, the
+        substituted error object lands under the MAIN seq ids — poll for it
+        so the consumer never hangs on a missing chunk."""
+        import time as _time
+
+        remaining = set(range(k))
+        deadline = _time.monotonic() + 600
+        while remaining:
+            got = None
+            for i in sorted(remaining):
+                b = self._server.try_take(f"{up}\x01x{i}", down)
+                if b is not None:
+                    got = (i, b)
+                    break
+            if got is None:
+                err = self._server.try_take(up, down)
+                if err is not None:
+                    self._take(up, down, err)  # raises for error frames
+                    raise RuntimeError(
+                        "unexpected non-error frame during chunked receive"
+                    )
+                if _time.monotonic() > deadline:
+                    raise TimeoutError("chunk-streamed receive stalled")
+                i = min(remaining)
+                try:
+                    b = self._server.get_data(f"{up}\x01x{i}", down, 0.25)
+                    got = (i, b)
+                except RuntimeError as e:
+                    if "timeout" in str(e):
+                        continue
+                    raise
+            remaining.discard(got[0])
+            yield got
 
     def get_data(self, src_party, upstream_seq_id, curr_seq_id) -> Future:
         up, down = str(upstream_seq_id), str(curr_seq_id)

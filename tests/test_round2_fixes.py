@@ -249,3 +249,34 @@ def test_striped_send_roundtrip():
     finally:
         send.stop()
         recv.stop()
+
+
+def test_chunk_streamed_send_roundtrip(monkeypatch):
+    """Tensor frames above 2x the stripe size travel as KIND_CHUNKED main
+    frame + sidecar chunk frames consumed in arrival order; content and a
+    trailing small frame must both survive."""
+    torch = pytest.importorskip("torch")
+    from rayfed_amd.proxy.xfer import xfer_available
+
+    if not xfer_available():
+        pytest.skip("C++ transport extension not built")
+    monkeypatch.setenv("RAYFED_SHM", "0")  # force the socket/payload route
+    from rayfed_amd.proxy.xfer import XferReceiverService, XferSenderService
+
+    addrs = make_addresses(["alice"])
+    cfg = GrpcCrossSiloMessageConfig.from_dict(
+        {"messages_max_size_in_bytes": 1 << 20}
+    )
+    recv = XferReceiverService(addrs["alice"], "alice", "j", cfg)
+    send = XferSenderService(addrs, "alice", "j", cfg)
+    try:
+        t = torch.arange(3 << 20, dtype=torch.int16)  # 6 MiB > 2x 1 MiB
+        obj = {"w": t, "tag": 5}
+        assert send.send("alice", obj, "500", "500").result(timeout=60)
+        out = recv.get_data("alice", "500", "500").result(timeout=60)
+        assert out["tag"] == 5 and torch.equal(out["w"], t)
+        assert send.send("alice", 99, "501", "501").result(timeout=30)
+        assert recv.get_data("alice", "501", "501").result(timeout=30) == 99
+    finally:
+        send.stop()
+        recv.stop()
