@@ -60,9 +60,86 @@ constexpr uint8_t kFlagDeferAck = 1;
 // the shared assembly buffer at `off` (no reassembly copy) and the last
 // stripe posts the completed body to the mailbox.
 constexpr uint8_t kFlagStripe = 2;
+// PINNED body: assemble this (striped) frame into hipHostMalloc'd memory
+// and deliver it to Python as a zero-copy view (wait_view) — the consumer
+// H2Ds straight out of it, no bytes copy, no pinned staging bounce.
+constexpr uint8_t kFlagPinned = 4;
 // Reserved seq id of the init-time readiness barrier (constants.PING_SEQ_ID):
 // acked without parking so repeated pings never leak mailbox slots.
 constexpr const char* kPingSeqId = "ping";
+
+// hipHostMalloc/hipHostFree, linked from amdhip64; declared here so this
+// translation unit needs no HIP headers (plain g++ build).
+extern "C" int hipHostMalloc(void** ptr, size_t size, unsigned int flags);
+extern "C" int hipHostFree(void* ptr);
+
+// Pooled pinned host buffers (power-of-two classes).  hipHostMalloc costs
+// ~10 ms/100 MiB (page-locking), so chunk-streamed receives reuse buffers
+// across chunks and transfers.  Falls back to plain malloc when no GPU
+// runtime is live (CPU boxes) — the consumer then stages through its own
+// pinned pool as before.
+struct PinnedBuf {
+  char* p = nullptr;
+  size_t cap = 0;
+  size_t len = 0;  // valid bytes of the current body
+  bool pinned = false;
+  ~PinnedBuf();
+};
+
+class PinnedPool {
+ public:
+  static PinnedPool& inst() {
+    static PinnedPool pool;
+    return pool;
+  }
+
+  std::shared_ptr<PinnedBuf> acquire(size_t n) {
+    size_t cls = 1ull << (64 - __builtin_clzll(std::max<size_t>(n, 1 << 16) - 1));
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto& bucket = free_[cls];
+      if (!bucket.empty()) {
+        auto b = bucket.back();
+        bucket.pop_back();
+        b->len = n;
+        return b;
+      }
+    }
+    auto b = std::make_shared<PinnedBuf>();
+    void* p = nullptr;
+    if (hipHostMalloc(&p, cls, 0) == 0 && p) {
+      b->p = (char*)p;
+      b->pinned = true;
+    } else {
+      b->p = (char*)malloc(cls);
+      if (!b->p) throw std::bad_alloc();
+      b->pinned = false;
+    }
+    b->cap = cls;
+    b->len = n;
+    return b;
+  }
+
+  void release(std::shared_ptr<PinnedBuf> b) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto& bucket = free_[b->cap];
+    size_t limit = b->cap >= (64u << 20) ? 24 : 8;
+    if (bucket.size() < limit) bucket.push_back(std::move(b));
+    // else: shared_ptr drops it; ~PinnedBuf frees
+  }
+
+ private:
+  std::mutex mu_;
+  std::unordered_map<size_t, std::vector<std::shared_ptr<PinnedBuf>>> free_;
+};
+
+PinnedBuf::~PinnedBuf() {
+  if (!p) return;
+  if (pinned)
+    hipHostFree(p);
+  else
+    free(p);
+}
 
 // ---------------------------------------------------------------- utilities
 static void write_all(int fd, const char* data, size_t n) {
@@ -220,6 +297,28 @@ static SSL_CTX* make_client_ctx(const std::string& ca, const std::string& cert,
   return ctx;
 }
 
+// Zero-copy Python view over a completed pinned body; returns the buffer
+// to the pool when garbage-collected (after the consumer's H2D completes —
+// the exporter stays alive while memoryviews over it exist).
+class BodyView {
+ public:
+  explicit BodyView(std::shared_ptr<PinnedBuf> b) : buf_(std::move(b)) {}
+  ~BodyView() {
+    if (buf_) {
+      try {
+        PinnedPool::inst().release(std::move(buf_));
+      } catch (...) {
+      }
+    }
+  }
+  char* data() const { return buf_->p; }
+  size_t size() const { return buf_->len; }
+  bool pinned() const { return buf_->pinned; }
+
+ private:
+  std::shared_ptr<PinnedBuf> buf_;
+};
+
 // ------------------------------------------------------------------- server
 class XferServer {
  public:
@@ -289,9 +388,10 @@ class XferServer {
       std::lock_guard<std::mutex> lk(tok_mu_);
       tokens_.clear();
     }
-    // Unblock any waiting get_data.
+    // Unblock any waiting get_data / wait_view.
     std::lock_guard<std::mutex> lk(mail_mu_);
     stopped_ = true;
+    pinned_mail_.clear();
     mail_cv_.notify_all();
   }
 
@@ -375,6 +475,47 @@ class XferServer {
     mail_cv_.notify_all();
   }
 
+  // Pinned-body mail: blocking / non-blocking fetch of a completed
+  // kFlagPinned body as a zero-copy BodyView.
+  py::object wait_view(const std::string& up, const std::string& down,
+                       double timeout_s) {
+    std::string key = up + '\x00' + down;
+    std::shared_ptr<PinnedBuf> b;
+    {
+      py::gil_scoped_release release;
+      std::unique_lock<std::mutex> lk(mail_mu_);
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::duration<double>(timeout_s);
+      while (true) {
+        auto it = pinned_mail_.find(key);
+        if (it != pinned_mail_.end()) {
+          b = std::move(it->second);
+          pinned_mail_.erase(it);
+          break;
+        }
+        if (stopped_) throw std::runtime_error("server stopped");
+        if (mail_cv_.wait_until(lk, deadline) == std::cv_status::timeout)
+          throw std::runtime_error("wait_view timeout");
+      }
+    }
+    return py::cast(new BodyView(std::move(b)),
+                    py::return_value_policy::take_ownership);
+  }
+
+  py::object try_view(const std::string& up, const std::string& down) {
+    std::string key = up + '\x00' + down;
+    std::shared_ptr<PinnedBuf> b;
+    {
+      std::lock_guard<std::mutex> lk(mail_mu_);
+      auto it = pinned_mail_.find(key);
+      if (it == pinned_mail_.end()) return py::none();
+      b = std::move(it->second);
+      pinned_mail_.erase(it);
+    }
+    return py::cast(new BodyView(std::move(b)),
+                    py::return_value_policy::take_ownership);
+  }
+
   // Non-blocking probe for the fast path.
   std::optional<py::bytes> try_take(const std::string& up,
                                     const std::string& down) {
@@ -405,15 +546,23 @@ class XferServer {
 
   // In-flight striped body, filled by several connection threads at once.
   // `got` counts bytes landed; the stripe that completes the byte count
-  // moves the body to the mailbox.  (A retried frame after a partial
-  // failure recreates the slot only if its geometry changed; a stale mix
-  // is caught by the payload checksum downstream.)
+  // moves the body to the mailbox (or, for kFlagPinned frames, to the
+  // pinned-view mail).  (A retried frame after a partial failure recreates
+  // the slot only if its geometry changed; a stale mix is caught by the
+  // payload checksum downstream.)
   struct Assembly {
-    std::string data;
+    std::string data;                 // plain bodies
+    std::shared_ptr<PinnedBuf> pbuf;  // kFlagPinned bodies
     std::atomic<uint64_t> got{0};
     uint64_t total;
     uint32_t cnt;
-    Assembly(uint64_t t, uint32_t c) : total(t), cnt(c) { data.resize(t); }
+    Assembly(uint64_t t, uint32_t c, bool pinned) : total(t), cnt(c) {
+      if (pinned)
+        pbuf = PinnedPool::inst().acquire(t);
+      else
+        data.resize(t);
+    }
+    char* base() { return pbuf ? pbuf->p : data.data(); }
   };
 
   static bool drain(Stream& st, uint64_t n) {
@@ -513,18 +662,21 @@ class XferServer {
           result = "JobName mis-match: expected " + job_ + ", got " + job;
         } else {
           std::string key = up + '\x00' + down;
+          const bool pinned = (flags & kFlagPinned) != 0;
           std::shared_ptr<Assembly> asmb;
           {
             std::lock_guard<std::mutex> lk(asm_mu_);
             auto& slot = asm_[key];
             if (!slot || slot->total != body_total || slot->cnt != cnt)
-              slot = std::make_shared<Assembly>(body_total, cnt);
+              slot = std::make_shared<Assembly>(body_total, cnt, pinned);
             asmb = slot;
           }
           // Payload streams DIRECTLY into the assembly buffer at its
           // offset — stripes of one frame write disjoint ranges from
-          // their own connection threads, no reassembly copy.
-          if (!st.read_all(asmb->data.data() + s_off, payload_len)) break;
+          // their own connection threads, no reassembly copy.  For pinned
+          // bodies the buffer is hipHostMalloc'd, so the socket reads land
+          // in DMA-able memory.
+          if (!st.read_all(asmb->base() + s_off, payload_len)) break;
           if (asmb->got.fetch_add(payload_len) + payload_len == body_total) {
             {
               std::lock_guard<std::mutex> lk(asm_mu_);
@@ -532,7 +684,10 @@ class XferServer {
             }
             recv_count_.fetch_add(1);
             std::lock_guard<std::mutex> lk(mail_mu_);
-            mail_[key] = std::move(asmb->data);
+            if (asmb->pbuf)
+              pinned_mail_[key] = std::move(asmb->pbuf);
+            else
+              mail_[key] = std::move(asmb->data);
             mail_cv_.notify_all();
           }
         }
@@ -611,6 +766,7 @@ class XferServer {
   std::map<std::string, std::string> mail_;
   std::mutex asm_mu_;
   std::unordered_map<std::string, std::shared_ptr<Assembly>> asm_;
+  std::unordered_map<std::string, std::shared_ptr<PinnedBuf>> pinned_mail_;
   std::atomic<uint64_t> recv_count_{0};
   std::mutex tok_mu_;
   std::unordered_map<uint64_t, std::pair<std::shared_ptr<SrvConn>, uint64_t>>
@@ -660,9 +816,9 @@ class XferClient {
                    const std::string& down,
                    const std::vector<std::pair<const char*, size_t>>& views,
                    uint64_t body_len, int stripes, double timeout_s,
-                   std::string* result_out) {
+                   std::string* result_out, bool pinned = false) {
     std::string names;
-    names.push_back((char)(kFlagStripe));
+    names.push_back((char)(kFlagStripe | (pinned ? kFlagPinned : 0)));
     names.push_back((char)job_.size());
     names.push_back((char)up.size());
     names.push_back((char)down.size());
@@ -799,7 +955,7 @@ class XferClient {
   int send(const std::string& host, int port, const std::string& up,
            const std::string& down, std::vector<py::buffer> parts,
            bool defer_ack, double timeout_s, std::string* result_out,
-           int stripes = 1) {
+           int stripes = 1, bool pinned = false) {
     // Collect buffer pointers under the GIL.
     std::vector<std::pair<const char*, size_t>> views;
     views.reserve(parts.size());
@@ -817,7 +973,7 @@ class XferClient {
     if (stripes > 1 && !defer_ack && body_len > 0) {
       py::gil_scoped_release release;
       return send_striped(host, port, up, down, views, body_len, stripes,
-                          timeout_s, result_out);
+                          timeout_s, result_out, pinned);
     }
 
     uint8_t flags = defer_ack ? kFlagDeferAck : 0;
@@ -1222,6 +1378,14 @@ class XferClient {
 
 PYBIND11_MODULE(_xfer, m) {
   m.doc() = "rayfed_amd C++ transport core (plaintext cross-silo hot path)";
+  py::class_<BodyView>(m, "BodyView", py::buffer_protocol())
+      .def_buffer([](BodyView& v) {
+        return py::buffer_info(v.data(), sizeof(uint8_t),
+                               py::format_descriptor<uint8_t>::format(), 1,
+                               {v.size()}, {sizeof(uint8_t)});
+      })
+      .def("__len__", &BodyView::size)
+      .def_property_readonly("pinned", &BodyView::pinned);
   py::class_<XferServer>(m, "XferServer")
       .def(py::init<int, std::string, std::string, std::string, std::string>(),
            py::arg("port"), py::arg("job_name"), py::arg("tls_cert") = "",
@@ -1235,6 +1399,9 @@ PYBIND11_MODULE(_xfer, m) {
            py::arg("code"), py::arg("result") = std::string(),
            py::call_guard<py::gil_scoped_release>())
       .def("post", &XferServer::post, py::call_guard<py::gil_scoped_release>())
+      .def("wait_view", &XferServer::wait_view, py::arg("up"),
+           py::arg("down"), py::arg("timeout_s") = 60.0)
+      .def("try_view", &XferServer::try_view)
       .def("try_take", &XferServer::try_take)
       .def_property_readonly("received_op_count",
                              &XferServer::received_op_count);
@@ -1249,16 +1416,17 @@ PYBIND11_MODULE(_xfer, m) {
           [](XferClient& c, const std::string& host, int port,
              const std::string& up, const std::string& down,
              std::vector<py::buffer> parts, bool defer_ack,
-             double timeout_s, int stripes) {
+             double timeout_s, int stripes, bool pinned) {
             std::string result;
             int code =
                 c.send(host, port, up, down, std::move(parts), defer_ack,
-                       timeout_s, &result, stripes);
+                       timeout_s, &result, stripes, pinned);
             return py::make_tuple(code, result);
           },
           py::arg("host"), py::arg("port"), py::arg("up"), py::arg("down"),
           py::arg("parts"), py::arg("defer_ack") = false,
-          py::arg("timeout_s") = 60.0, py::arg("stripes") = 1)
+          py::arg("timeout_s") = 60.0, py::arg("stripes") = 1,
+          py::arg("pinned") = false)
       .def("send_async", &XferClient::send_async, py::arg("host"),
            py::arg("port"), py::arg("up"), py::arg("down"), py::arg("parts"),
            py::arg("defer_ack") = false)

@@ -670,26 +670,38 @@ class GpuDataPlane:
 
     # -- streamed H2D (chunked socket lane) ----------------------------------
     def h2d_copy(self, dst_flat_u8: "torch.Tensor", dst_off: int, data):
-        """Stage ``data`` through a pooled pinned buffer and H2D it into
-        ``dst_flat_u8[dst_off:]`` asynchronously on the copy stream.  The
-        CPU memcpy of the NEXT chunk overlaps this chunk's DMA.  Returns
-        (event, pinned) for :meth:`finish_h2d`."""
+        """H2D ``data`` into ``dst_flat_u8[dst_off:]`` asynchronously on the
+        copy stream.  A body already in hipHostMalloc'd memory (the C++
+        transport's BodyView, .pinned True) DMAs zero-copy; anything else
+        stages through a pooled pinned buffer — the CPU memcpy of the NEXT
+        chunk overlaps this chunk's DMA.  Returns a token for
+        :meth:`finish_h2d` (which also keeps the source alive until the
+        DMA completes)."""
         self._bind_device()
-        n = len(data)
+        mv = memoryview(data).cast("B")
+        n = len(mv)
+        if getattr(data, "pinned", False):
+            src = torch.frombuffer(mv, dtype=torch.uint8)
+            with torch.cuda.stream(self._copy_stream):
+                dst_flat_u8[dst_off : dst_off + n].copy_(src, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record(self._copy_stream)
+            return ev, None, data  # hold the view until the DMA is done
         pinned = self._get_buf(self._pinned, n, pin=True)
-        pinned[:n].numpy()[:] = memoryview(data).cast("B")
+        pinned[:n].numpy()[:] = mv
         with torch.cuda.stream(self._copy_stream):
             dst_flat_u8[dst_off : dst_off + n].copy_(
                 pinned[:n], non_blocking=True
             )
             ev = torch.cuda.Event()
             ev.record(self._copy_stream)
-        return ev, pinned
+        return ev, pinned, None
 
     def finish_h2d(self, pend) -> None:
-        for ev, pinned in pend:
+        for ev, pinned, _keepalive in pend:
             ev.synchronize()
-            self._put_buf(self._pinned, pinned)
+            if pinned is not None:
+                self._put_buf(self._pinned, pinned)
 
     def device_crc32(self, flat_u8: "torch.Tensor") -> int:
         out = self._ext.crc32_async(flat_u8)

@@ -68,6 +68,12 @@ def _prestart_pool(pool: ThreadPoolExecutor, n: int) -> None:
         pass
 
 
+def _xfer_debug() -> bool:
+    import os
+
+    return os.environ.get("RAYFED_XFER_DEBUG") == "1"
+
+
 def _coerce_config(proxy_config):
     if proxy_config is not None and not isinstance(
         proxy_config, fed_config.CrossSiloMessageConfig
@@ -396,9 +402,21 @@ class XferSenderService:
                     # Chunk-streamed tensor frame: the receiver H2Ds each
                     # chunk as it lands, overlapping consume with arrival
                     # (a whole-frame send serializes network then consume).
+                    if _xfer_debug():
+                        import sys as _sys
+
+                        print(f"[xfer-tx] chunked send {up}/{down} "
+                              f"payload={payload_len>>20}MiB t={time.monotonic():.3f}",
+                              file=_sys.stderr, flush=True)
                     code, result = self._send_chunked(
                         host, int(port), up, down, body_parts, payload_len
                     )
+                    if _xfer_debug():
+                        import sys as _sys
+
+                        print(f"[xfer-tx] chunked done {up}/{down} "
+                              f"t={time.monotonic():.3f}",
+                              file=_sys.stderr, flush=True)
                 else:
                     bulk = defer_ack or nbytes > (1 << 20)
                     stripes = 1
@@ -444,7 +462,7 @@ class XferSenderService:
             # starve under concurrent big sends.
             futs.append(self._chunk_pool.submit(
                 self._send_with_retry, self._client_bulk, host, port,
-                f"{up}\x01x{i}", down, sub, False, 4,
+                f"{up}\x01x{i}", down, sub, False, 4, True,
             ))
         meta = {
             "job": self._job_name, "up": up, "down": down,
@@ -464,7 +482,7 @@ class XferSenderService:
         return code, result
 
     def _send_with_retry(self, client, host, port, up, down, parts, defer_ack,
-                         stripes: int = 1):
+                         stripes: int = 1, pinned: bool = False):
         deadline = time.monotonic() + self._timeout_s
         backoff = self._retry.initial_backoff
         attempt = 0
@@ -474,7 +492,7 @@ class XferSenderService:
                 remaining = max(0.001, deadline - time.monotonic())
                 return client.send(
                     host, port, up, down, parts, defer_ack, remaining,
-                    stripes,
+                    stripes, pinned,
                 )
             except RuntimeError as e:
                 now = time.monotonic()
@@ -619,7 +637,7 @@ class XferReceiverService:
         while remaining:
             got = None
             for i in sorted(remaining):
-                b = self._server.try_take(f"{up}\x01x{i}", down)
+                b = self._server.try_view(f"{up}\x01x{i}", down)
                 if b is not None:
                     got = (i, b)
                     break
@@ -634,13 +652,19 @@ class XferReceiverService:
                     raise TimeoutError("chunk-streamed receive stalled")
                 i = min(remaining)
                 try:
-                    b = self._server.get_data(f"{up}\x01x{i}", down, 0.25)
+                    b = self._server.wait_view(f"{up}\x01x{i}", down, 0.25)
                     got = (i, b)
                 except RuntimeError as e:
                     if "timeout" in str(e):
                         continue
                     raise
             remaining.discard(got[0])
+            if _xfer_debug():
+                import sys as _sys
+
+                print(f"[xfer-rx] chunk {got[0]} ({len(got[1])>>20}MiB) "
+                      f"t={_time.monotonic():.3f}",
+                      file=_sys.stderr, flush=True)
             yield got
 
     def get_data(self, src_party, upstream_seq_id, curr_seq_id) -> Future:

@@ -31,7 +31,12 @@ setup(
             name="rayfed_amd._xfer",
             sources=["csrc/xfer_core.cpp"],
             extra_compile_args=["-O3", "-std=c++17", "-pthread"],
-            libraries=["ssl", "crypto"],
+            # amdhip64: striped chunk bodies assemble into hipHostMalloc'd
+            # (pinned) buffers so the consumer H2Ds them zero-copy; falls
+            # back to plain malloc when no GPU is present.
+            libraries=["ssl", "crypto", "amdhip64"],
+            library_dirs=["/opt/rocm/lib"],
+            include_dirs=["/opt/rocm/include"],
         ),
     ],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
