@@ -177,6 +177,17 @@ def init(
         pool = DeviceWorkerPool(
             devices=list(party_gpus),
             with_party_group=len(party_gpus) > 1,
+            kv_snapshot={
+                "job_name": job_name,
+                "entries": {
+                    constants.KEY_OF_CLUSTER_CONFIG: kv.get(
+                        constants.KEY_OF_CLUSTER_CONFIG
+                    ),
+                    constants.KEY_OF_JOB_CONFIG: kv.get(
+                        constants.KEY_OF_JOB_CONFIG
+                    ),
+                },
+            },
         )
         ctx.get_executor().attach_worker_pool(pool)
 

@@ -48,12 +48,23 @@ class RemoteHandle:
 
 
 def _worker_main(worker_id: int, device: Optional[int], conn,
-                 group_cfg: Optional[dict]):
+                 group_cfg: Optional[dict],
+                 kv_snapshot: Optional[dict] = None):
     """Worker process loop: execute tasks, keep an object table."""
     import torch
 
     if device is not None and torch.cuda.is_available():
         torch.cuda.set_device(device)
+    if kv_snapshot:
+        # Seed this process's internal KV with the driver's job/cluster
+        # config so tasks can read fed_config.get_cluster_config() etc.
+        # (the reference gets this via Ray's GCS KV — compatible_utils.py).
+        from rayfed_amd._private import kv as kv_mod
+
+        k = kv_mod._init_internal_kv(kv_snapshot["job_name"])
+        for key, value in kv_snapshot["entries"].items():
+            if value is not None:
+                k.put(key, value)
     if group_cfg is not None:
         from rayfed_amd.parallel.group import init_party_group
 
@@ -133,6 +144,7 @@ class DeviceWorkerPool:
         master_port: Optional[int] = None,
         backend: Optional[str] = None,
         start_method: str = "spawn",
+        kv_snapshot: Optional[dict] = None,
     ):
         ctx = mp.get_context(start_method)
         self._task_counter = itertools.count(1)
@@ -159,7 +171,7 @@ class DeviceWorkerPool:
             parent, child = ctx.Pipe()
             proc = ctx.Process(
                 target=_worker_main,
-                args=(wid, dev, child, group_cfg),
+                args=(wid, dev, child, group_cfg, kv_snapshot),
                 name=f"rayfed-worker-{wid}",
                 daemon=True,
             )
