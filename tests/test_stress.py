@@ -171,6 +171,8 @@ def test_receiver_crash_mid_push_escalates():
     _mp = multiprocessing.get_context("fork")
 
     def alice_main():
+        import time as _time
+
         import rayfed_amd as _fed
 
         def handler(err):
@@ -181,32 +183,39 @@ def test_receiver_crash_mid_push_escalates():
             sending_failure_handler=handler,
             config={"cross_silo_comm": {
                 "exit_on_sending_failure": True,
-                "timeout_in_ms": 5000,
+                "timeout_in_ms": 4000,
                 "grpc_retry_policy": {"maxAttempts": 2,
                                       "initialBackoff": "0.2s"},
             }},
         )
+
         @_fed.remote
         def ping_round(x):
             return x + 1
 
-        o = ping_round.party("alice").remote(1)
-        r = ping_round.party("bob").remote(o)
-        assert _fed.get(r) == 3
-
         @_fed.remote
-        def big():
-            return torch.ones(1 << 21)
+        def big(i):
+            return torch.ones(1 << 21)  # 8 MiB — rides the defer-ack lane
 
         @_fed.remote
         def sink(t):
             return int(t.numel())
 
-        o2 = big.party("alice").remote()
-        sink.party("bob").remote(o2)
-        import time as _time
-
-        _time.sleep(30)
+        o = ping_round.party("alice").remote(1)
+        r = ping_round.party("bob").remote(o)
+        assert _fed.get(r) == 3
+        # Handshake: bob dies only after this value reaches him, so the
+        # crash always lands mid-job — never before alice's round-1 recv
+        # completes (a peer dying pre-broadcast hangs the recv by design,
+        # exactly as the reference would).
+        go = ping_round.party("alice").remote(10)
+        _fed.get(go)
+        # Keep pushing until a send hits the dead peer; exit_on_sending_
+        # failure then SIGINTs this process out of the loop.
+        for i in range(20):
+            o2 = big.party("alice").remote(i)
+            sink.party("bob").remote(o2)
+            _time.sleep(0.75)
         raise RuntimeError("sending failure never escalated")
 
     def bob_main():
@@ -220,9 +229,19 @@ def test_receiver_crash_mid_push_escalates():
         def ping_round(x):
             return x + 1
 
+        @_fed.remote
+        def big(i):
+            return torch.ones(1 << 21)
+
+        @_fed.remote
+        def sink(t):
+            return int(t.numel())
+
         o = ping_round.party("alice").remote(1)
         r = ping_round.party("bob").remote(o)
         assert _fed.get(r) == 3
+        go = ping_round.party("alice").remote(10)
+        assert _fed.get(go) == 11  # alice's round-1 view is complete
         _os._exit(3)
 
     pa = _mp.Process(target=alice_main)
