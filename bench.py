@@ -306,7 +306,10 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
     dev = f"cuda:{device}" if use_gpu else "cpu"
     if use_gpu:
         torch.cuda.set_device(device)
+    # Zero-copy receive: the combine reads the peer's IPC slabs directly
+    # (fused combine+verify kernel) instead of materializing a copy.
     fed.init(addresses=addresses, party=party, job_name=job_name,
+             config={"gpu_data_plane": {"lazy_ipc": True}},
              logging_level="warning")
 
     dtype = torch.bfloat16

@@ -182,6 +182,26 @@ __global__ void crc32_kernel(const uint8_t* __restrict__ data,
   block_xor_out(contrib_tail, &out[1], lds_scratch);
 }
 
+__device__ __forceinline__ float bf16_to_f32(uint16_t h) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.u = (uint32_t)h << 16;
+  return c.f;
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+  union {
+    uint32_t u;
+    float f;
+  } c;
+  c.f = f;
+  // round-to-nearest-even
+  uint32_t r = c.u + 0x7FFFu + ((c.u >> 16) & 1u);
+  return (uint16_t)(r >> 16);
+}
+
 // ---------------------------------------------------------------------------
 // hash64 — memory-rate 64-bit integrity hash for the device-IPC lane.
 //
@@ -265,6 +285,94 @@ __global__ void hash64_kernel(const unsigned long long* __restrict__ words,
   block_xor_out64(v, out, lds);
 }
 
+// ---------------------------------------------------------------------------
+// Fused FedAvg combine + integrity hash (zero-copy receive path).
+//
+// out[i] = wa*local[i] + wb*peer[i] (bf16, fp32 math) while computing the
+// hash64 of the PEER bytes in the same pass — the receiver combines
+// directly from the sender's IPC slab, never materializing a copy and
+// never re-reading it for the verify.  The thread/word mapping is
+// IDENTICAL to hash64_kernel (fixed kHashLanes slots), so the sender's
+// per-slab hash values verify unchanged.
+// ---------------------------------------------------------------------------
+__global__ void fedavg_combine_hash_kernel(
+    uint16_t* __restrict__ out, const uint16_t* __restrict__ local,
+    const unsigned long long* __restrict__ peer_words,
+    unsigned long long n_words, const uint8_t* __restrict__ peer_tail,
+    uint32_t tail_len, unsigned long long nbytes, float wa, float wb,
+    unsigned long long* __restrict__ hash_out) {
+  __shared__ unsigned long long lds[kBlock / kWave];
+  const uint32_t lane = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long slots = 4ull * kHashLanes;
+  const unsigned long long* lw =
+      reinterpret_cast<const unsigned long long*>(local);
+  unsigned long long h[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k)
+    h[k] = (kFnvOff ^ (unsigned long long)(lane * 4u + k)) * kFnvP;
+  for (unsigned long long base = (unsigned long long)lane * 4ull;
+       base < n_words; base += slots) {
+    if (base + 4 <= n_words) {
+      unsigned long long pw[4], ow[4];
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        pw[k] = peer_words[base + k];
+        h[k] = (h[k] ^ pw[k]) * kFnvP;
+      }
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        const unsigned long long lv = lw[base + k];
+        unsigned long long o = 0;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float p = bf16_to_f32((uint16_t)(pw[k] >> (16 * e)));
+          const float l = bf16_to_f32((uint16_t)(lv >> (16 * e)));
+          o |= (unsigned long long)f32_to_bf16(wa * l + wb * p) << (16 * e);
+        }
+        ow[k] = o;
+      }
+      unsigned long long* outw = reinterpret_cast<unsigned long long*>(out);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) outw[base + k] = ow[k];
+    } else {
+      for (unsigned long long i = base; i < n_words; ++i) {
+        const unsigned long long pw = peer_words[i];
+        const unsigned long long lv = lw[i];
+        h[i - base] = (h[i - base] ^ pw) * kFnvP;
+        unsigned long long o = 0;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float p = bf16_to_f32((uint16_t)(pw >> (16 * e)));
+          const float l = bf16_to_f32((uint16_t)(lv >> (16 * e)));
+          o |= (unsigned long long)f32_to_bf16(wa * l + wb * p) << (16 * e);
+        }
+        reinterpret_cast<unsigned long long*>(out)[i] = o;
+      }
+    }
+  }
+  unsigned long long hl =
+      ((((h[0] * kFnvP ^ h[1]) * kFnvP ^ h[2]) * kFnvP ^ h[3]) * kFnvP);
+  unsigned long long v = fmix64(hl);
+  if (lane == 0) {
+    if (tail_len) {
+      unsigned long long tw = 0;
+      for (uint32_t i = 0; i < tail_len; ++i)
+        tw |= (unsigned long long)peer_tail[i] << (8 * i);
+      v ^= fmix64((kFnvOff ^ tw) * kFnvP);
+      // Combine the tail elements too (bf16 => tail_len is even).
+      const unsigned long long e0 = n_words * 4;
+      for (uint32_t e = 0; e < tail_len / 2; ++e) {
+        const float p = bf16_to_f32(
+            (uint16_t)(peer_tail[2 * e] | (peer_tail[2 * e + 1] << 8)));
+        const float l = bf16_to_f32(local[e0 + e]);
+        out[e0 + e] = f32_to_bf16(wa * l + wb * p);
+      }
+    }
+    v ^= fmix64((nbytes * kFnvP) ^ kFnvOff);
+  }
+  block_xor_out64(v, hash_out, lds);
+}
+
 struct TailParam {
   uint32_t m[32];       // shift matrix for the tail length
   uint32_t final_xor;   // shift_len(0xFFFFFFFF) ^ 0xFFFFFFFF
@@ -282,26 +390,6 @@ __global__ void crc_finalize_kernel(uint32_t* out, TailParam p) {
 // ---------------------------------------------------------------------------
 // fp8 wire cast (OCP e4m3), fused with CRC over the produced bytes
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ float bf16_to_f32(uint16_t h) {
-  union {
-    uint32_t u;
-    float f;
-  } c;
-  c.u = (uint32_t)h << 16;
-  return c.f;
-}
-
-__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
-  union {
-    uint32_t u;
-    float f;
-  } c;
-  c.f = f;
-  // round-to-nearest-even
-  uint32_t r = c.u + 0x7FFFu + ((c.u >> 16) & 1u);
-  return (uint16_t)(r >> 16);
-}
-
 __device__ __forceinline__ uint8_t f32_to_fp8(float f) {
   __hip_fp8_e4m3 v(f);
   return v.__x;
@@ -734,6 +822,43 @@ void unpack_fp8_async(torch::Tensor src, torch::Tensor dst) {
                      reinterpret_cast<uint16_t*>(dst.data_ptr()), n);
 }
 
+torch::Tensor fedavg_combine_hash_async(torch::Tensor out,
+                                        torch::Tensor local,
+                                        torch::Tensor peer_bytes, double wa,
+                                        double wb) {
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() &&
+                  out.dtype() == torch::kBFloat16,
+              "combine_hash: out must be contiguous CUDA bf16");
+  TORCH_CHECK(local.is_cuda() && local.is_contiguous() &&
+                  local.dtype() == torch::kBFloat16 &&
+                  local.numel() == out.numel(),
+              "combine_hash: local mismatch");
+  TORCH_CHECK(peer_bytes.is_cuda() && peer_bytes.is_contiguous() &&
+                  peer_bytes.dtype() == torch::kUInt8 &&
+                  peer_bytes.numel() == out.numel() * 2,
+              "combine_hash: peer byte count mismatch");
+  const unsigned long long nbytes = peer_bytes.numel();
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(peer_bytes.data_ptr()) & 7u) == 0 &&
+                  (reinterpret_cast<uintptr_t>(local.data_ptr()) & 7u) == 0 &&
+                  (reinterpret_cast<uintptr_t>(out.data_ptr()) & 7u) == 0,
+              "combine_hash requires 8-byte alignment");
+  auto hash_out = torch::zeros(
+      {1}, torch::dtype(torch::kInt64).device(out.device()));
+  const unsigned long long n_words = nbytes / 8;
+  const uint32_t tail_len = (uint32_t)(nbytes % 8);
+  auto stream = at::hip::getCurrentHIPStream();
+  const uint8_t* p = static_cast<const uint8_t*>(peer_bytes.data_ptr());
+  hipLaunchKernelGGL(fedavg_combine_hash_kernel, dim3(kHashLanes / kBlock),
+                     dim3(kBlock), 0, stream,
+                     reinterpret_cast<uint16_t*>(out.data_ptr()),
+                     reinterpret_cast<const uint16_t*>(local.data_ptr()),
+                     reinterpret_cast<const unsigned long long*>(p), n_words,
+                     p + n_words * 8, tail_len, nbytes, (float)wa, (float)wb,
+                     reinterpret_cast<unsigned long long*>(
+                         hash_out.data_ptr<int64_t>()));
+  return hash_out;
+}
+
 void fedavg_reduce_(torch::Tensor out, std::vector<torch::Tensor> inputs,
                     std::vector<double> weights) {
   TORCH_CHECK(!inputs.empty() && inputs.size() <= kMaxInputs,
@@ -899,6 +1024,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("unpack_fp8_async", &unpack_fp8_async, "fp8 e4m3 bytes -> bf16");
   m.def("fedavg_reduce_", &fedavg_reduce_,
         "out = sum_k w_k * in_k (bf16/f16/f32, fp32 accumulation)");
+  m.def("fedavg_combine_hash_async", &fedavg_combine_hash_async,
+        "out = wa*local + wb*peer_bytes(bf16) fused with the hash64 of the "
+        "peer bytes -> int64[1] device tensor (zero-copy IPC combine)");
   m.def("fedavg_reduce_mfma_", &fedavg_reduce_mfma_,
         "MFMA (v_mfma_f32_16x16x32_bf16) variant of the weighted combine");
   m.def("masked_add_", &masked_add_, "dst += mask ? src : 0");

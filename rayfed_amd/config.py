@@ -145,6 +145,12 @@ class GpuDataPlaneConfig:
     device_checksum: str = "fnv64"
     # Receive tensors straight back onto the GPU of the consuming party.
     place_on_gpu: bool = True
+    # Zero-copy receive for device-IPC tensors (C++ transport only): decode
+    # returns a LazyIpcTensor over the sender's slabs instead of a
+    # materialized copy; consumers combine/materialize from it and then
+    # release() it, which acks the sender (licensing slab reuse).  Opt-in:
+    # the object is not a torch.Tensor.
+    lazy_ipc: bool = False
     # Optional lossy wire compression for bf16 tensors: 'fp8e4m3' casts to
     # OCP fp8 on the wire (fused HIP cast+CRC kernel), halving bytes.
     wire_dtype: Optional[str] = None

@@ -65,6 +65,7 @@ class GpuDataPlane:
         self._ipc_pool: dict = {}  # size-class -> [slabs]
         self._own_ipc: dict = {}
         self._ipc_open_cache: dict = {}
+        self._pending_lazies: list = []  # LazyIpcTensors made by decode()
 
     def _bind_device(self):
         """hipSetDevice is per-THREAD: transport pool threads and the C++
@@ -668,6 +669,53 @@ class GpuDataPlane:
             crc = int(crc_out[2].item()) & 0xFFFFFFFF
         return seg, crc, lambda: pool.release(seg)
 
+    # -- zero-copy IPC receive (lazy_ipc) ------------------------------------
+    def combine_from_ipc(self, man, local_flat: "torch.Tensor",
+                         out_flat: "torch.Tensor", wa: float, wb: float):
+        """out = wa*local + wb*peer, reading the peer DIRECTLY from its IPC
+        slabs with the hash64 verify fused into the combine pass — no
+        materialized copy, no separate verify read.  bf16 only; the
+        manifest's checksums must be fnv64 (guaranteed when both sides run
+        this build with device_checksum='fnv64')."""
+        self._bind_device()
+        nbytes = man["nbytes"]
+        S = man["slab_bytes"]
+        handles = [bytes(h) for h in man["ipc_slabs"]]
+        crcs = man.get("ipc_crcs")
+        cks = man.get("cks") or []
+        if crcs is not None and any(k != "fnv64" for k in cks):
+            raise ValueError("combine_from_ipc requires fnv64 checksums")
+        local_u8 = local_flat.view(-1).view(torch.uint8)
+        out_u8 = out_flat.view(-1).view(torch.uint8)
+        hash_outs = []
+        for i, h in enumerate(handles):
+            lo, hi = i * S, min((i + 1) * S, nbytes)
+            src = self._ipc_src_view(h, hi - lo)[: hi - lo]
+            hash_outs.append(self._ext.fedavg_combine_hash_async(
+                out_u8[lo:hi].view(torch.bfloat16),
+                local_u8[lo:hi].view(torch.bfloat16),
+                src, wa, wb,
+            ))
+        torch.cuda.current_stream(self.device).synchronize()
+        if crcs is not None:
+            for i, ho in enumerate(hash_outs):
+                got = int(ho.item()) & 0xFFFFFFFFFFFFFFFF
+                if got != crcs[i]:
+                    raise ValueError(
+                        f"GPU tensor checksum mismatch (lazy combine, slab "
+                        f"{i}): expected {crcs[i]:#x}, got {got:#x}"
+                    )
+
+    # Created during decode when config.lazy_ipc is on; the transport's
+    # consume wires release() to the deferred ack.
+    def register_lazy(self, lazy: "LazyIpcTensor") -> None:
+        self._pending_lazies.append(lazy)
+
+    def pop_pending_lazies(self):
+        out = self._pending_lazies
+        self._pending_lazies = []
+        return out
+
     # -- streamed H2D (chunked socket lane) ----------------------------------
     def h2d_copy(self, dst_flat_u8: "torch.Tensor", dst_off: int, data):
         """H2D ``data`` into ``dst_flat_u8[dst_off:]`` asynchronously on the
@@ -791,6 +839,73 @@ class GpuDataPlane:
                     f"GPU tensor CRC mismatch: expected {crc_expect:#x}, got {crc:#x}"
                 )
         return out
+
+
+class LazyIpcTensor:
+    """Zero-copy handle over a peer's device-IPC slabs (config lazy_ipc).
+
+    Not a torch.Tensor: consumers either :meth:`materialize` it (D2D copy +
+    verify, like the eager path) or combine straight from the slabs via
+    ``rayfed_amd.parallel.fedavg.weighted_combine_`` (fused combine+verify
+    kernel).  :meth:`release` MUST be called when done — it acks the sender,
+    licensing slab reuse; dropping the object releases as a safety net.
+    """
+
+    def __init__(self, plane: GpuDataPlane, man: dict, dtype, shape):
+        self._plane = plane
+        self.man = man
+        self.dtype = dtype
+        self.shape = list(shape)
+        self._completer = None  # set by the transport's consume
+        self._released = False
+
+    @property
+    def nbytes(self) -> int:
+        return self.man["nbytes"]
+
+    def numel(self) -> int:
+        import math
+
+        return math.prod(self.shape) if self.shape else 1
+
+    def _attach_completer(self, fn) -> None:
+        self._completer = fn
+
+    def materialize(self) -> "torch.Tensor":
+        """D2D copy + verify into a regular tensor, then release."""
+        try:
+            return self._plane.unpack_from_ipc(self.man, self.dtype, self.shape)
+        finally:
+            self.release()
+
+    def combine_into(self, out, local, wa: float, wb: float) -> None:
+        """out = wa*local + wb*self — fused combine+verify, then release."""
+        try:
+            self._plane.combine_from_ipc(
+                self.man, local, out.view(-1), wa, wb
+            )
+        finally:
+            self.release()
+
+    def release(self) -> None:
+        if self._released:
+            return
+        self._released = True
+        if self._completer is not None:
+            try:
+                self._completer()
+            except Exception:  # noqa: BLE001 — ack path must not raise here
+                logger.warning("lazy IPC release failed", exc_info=True)
+
+    def __del__(self):  # safety net — never leave the sender unacked
+        try:
+            self.release()
+        except Exception:  # noqa: BLE001
+            pass
+
+    def __repr__(self):
+        return (f"LazyIpcTensor(shape={self.shape}, dtype={self.dtype}, "
+                f"released={self._released})")
 
 
 _plane: Optional[GpuDataPlane] = None

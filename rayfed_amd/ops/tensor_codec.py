@@ -440,10 +440,14 @@ def decode(
     payload: memoryview,
     gpu_plane=None,
     allowed_list: Optional[Dict] = None,
+    allow_lazy: bool = False,
 ) -> Any:
     """Inverse of :func:`encode`.  Tensors land on the GPU (via pinned H2D on
     a side stream) when ``gpu_plane`` is given and the manifest says the
-    source was device-resident; CRC32 is verified when present."""
+    source was device-resident; the device checksum is verified when
+    present.  ``allow_lazy`` (transports whose ack can be deferred until
+    consumption) + ``gpu_plane.config.lazy_ipc`` turn device-IPC tensors
+    into zero-copy :class:`LazyIpcTensor` handles instead of copies."""
     skel_len = extras["skel"]
     skeleton = payload[:skel_len]
     off = skel_len
@@ -469,6 +473,19 @@ def decode(
                     "received a device-IPC tensor but no GPU data plane is "
                     "attached (set RAYFED_IPC=0 on the sender for CPU peers)"
                 )
+            if (
+                allow_lazy
+                and gpu_plane.config.lazy_ipc
+                and not man.get("wire")
+            ):
+                from rayfed_amd.ops.gpu_plane import LazyIpcTensor
+
+                lazy = LazyIpcTensor(
+                    gpu_plane, man, _STR_TO_DTYPE[man["dtype"]], man["shape"]
+                )
+                gpu_plane.register_lazy(lazy)
+                tensors.append(lazy)
+                continue
             tensors.append(
                 gpu_plane.unpack_from_ipc(man, _STR_TO_DTYPE[man["dtype"]],
                                           man["shape"])

@@ -162,8 +162,30 @@ def weighted_combine_(
 
     On GPU this is the HIP ``fedavg_reduce_`` kernel (fp32 accumulation,
     16-byte lanes); on CPU a float32 torch reference with identical
-    numerics contract.
+    numerics contract.  A :class:`~rayfed_amd.ops.gpu_plane.LazyIpcTensor`
+    input (zero-copy receive, config lazy_ipc) dispatches the fused
+    combine+verify kernel reading the peer's slabs directly — supported for
+    the 2-input bf16 case (one lazy + one resident tensor).
     """
+    from rayfed_amd.ops.gpu_plane import LazyIpcTensor
+
+    lazy_idx = [i for i, t in enumerate(inputs) if isinstance(t, LazyIpcTensor)]
+    if lazy_idx:
+        if (
+            len(inputs) == 2
+            and len(lazy_idx) == 1
+            and out.is_cuda
+            and out.dtype == torch.bfloat16
+        ):
+            li = lazy_idx[0]
+            lazy, local = inputs[li], inputs[1 - li]
+            lazy.combine_into(out, local.view(-1), weights[1 - li], weights[li])
+            return out
+        # General case: materialize the lazies and fall through.
+        inputs = [
+            t.materialize() if isinstance(t, LazyIpcTensor) else t
+            for t in inputs
+        ]
     if out.is_cuda:
         from rayfed_amd.ops import _hip_loader
 

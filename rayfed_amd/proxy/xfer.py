@@ -592,16 +592,38 @@ class XferReceiverService:
                 memoryview(payload),
                 self.gpu_plane,
                 self._allowed_list,
+                allow_lazy=True,  # the ack can wait for lazy consumption
             )
         except Exception as e:  # noqa: BLE001
             logger.warning("xfer deferred consume failed: %r", e)
             self._server.complete(token, 500, f"consume failed: {e!r}")
             return
+        lazies = (
+            self.gpu_plane.pop_pending_lazies()
+            if self.gpu_plane is not None
+            else []
+        )
         with self._objs_lock:
             self._objs[(up, down)] = obj
             self._deferred_count += 1
         self._server.post(up, down, _OBJ_MARKER)
-        self._server.complete(token, 200)
+        if not lazies:
+            self._server.complete(token, 200)
+            return
+        # Zero-copy receive: the sender's slabs stay licensed to us until
+        # every lazy handle is released; the LAST release sends the ack.
+        remaining = [len(lazies)]
+        lock = threading.Lock()
+
+        def _one_done():
+            with lock:
+                remaining[0] -= 1
+                done = remaining[0] == 0
+            if done:
+                self._server.complete(token, 200)
+
+        for lz in lazies:
+            lz._attach_completer(_one_done)
 
     def _take(self, up: str, down: str, body: bytes):
         if bytes(body) == _OBJ_MARKER:

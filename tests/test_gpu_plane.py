@@ -489,3 +489,109 @@ def test_ipc_lane_fnv64_tamper_detected(plane):
             tensor_codec.decode(extras, memoryview(payload), plane, None)
     finally:
         tensor_codec.release_parts(extras)
+
+
+# ------------------------------------------------- zero-copy lazy IPC combine
+@needs_gpu
+def test_fedavg_combine_hash_matches_reference(ext):
+    """Fused combine+hash kernel: numerics vs an fp32 torch reference and
+    the hash vs the numpy hash64 reference (same slot mapping)."""
+    from rayfed_amd.ops.hash_ref import hash64_ref
+
+    torch.manual_seed(3)
+    for n in (16, 4096, (1 << 20) + 12):
+        a = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+        b = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+        out = torch.empty_like(a)
+        hv = ext.fedavg_combine_hash_async(
+            out, a, b.view(torch.uint8), 0.25, 0.75
+        )
+        torch.cuda.synchronize()
+        ref = (0.25 * a.float() + 0.75 * b.float()).to(torch.bfloat16)
+        assert torch.equal(out, ref), f"n={n}"
+        got = int(hv.item()) & 0xFFFFFFFFFFFFFFFF
+        expect = hash64_ref(b.view(torch.uint8).cpu().numpy().tobytes())
+        assert got == expect, f"n={n}"
+
+
+@needs_gpu
+def test_lazy_ipc_combine_roundtrip():
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane, LazyIpcTensor
+    from rayfed_amd.parallel.fedavg import weighted_combine_
+
+    plane = GpuDataPlane(GpuDataPlaneConfig(lazy_ipc=True))
+    torch.manual_seed(4)
+    n = 3 << 20
+    peer = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+    local = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+    extras, parts = tensor_codec.encode(peer, plane, shm=True)
+    payload = b"".join(bytes(p) for p in parts)
+    obj = tensor_codec.decode(
+        extras, memoryview(payload), plane, None, allow_lazy=True
+    )
+    assert isinstance(obj, LazyIpcTensor)
+    lazies = plane.pop_pending_lazies()
+    assert lazies == [obj]
+    released = []
+    obj._attach_completer(lambda: released.append(1))
+    out = torch.empty_like(local)
+    weighted_combine_(out, [local, obj], [0.5, 0.5])
+    assert released == [1], "combine must release (ack) the lazy handle"
+    ref = (0.5 * local.float() + 0.5 * peer.float()).to(torch.bfloat16)
+    assert torch.equal(out, ref)
+    tensor_codec.release_parts(extras)
+
+
+@needs_gpu
+def test_lazy_ipc_tamper_detected():
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+    from rayfed_amd.parallel.fedavg import weighted_combine_
+
+    plane = GpuDataPlane(GpuDataPlaneConfig(lazy_ipc=True))
+    peer = torch.randn(1 << 20, dtype=torch.bfloat16, device="cuda")
+    local = torch.randn(1 << 20, dtype=torch.bfloat16, device="cuda")
+    extras, parts = tensor_codec.encode(peer, plane, shm=True)
+    man = extras["tensors"][0]
+    handle = bytes(man["ipc_slabs"][0])
+    slab = plane._own_ipc[handle]
+    slab[3][4242] ^= 0xFF  # corrupt the slab AFTER the pack hashed it
+    payload = b"".join(bytes(p) for p in parts)
+    obj = tensor_codec.decode(
+        extras, memoryview(payload), plane, None, allow_lazy=True
+    )
+    plane.pop_pending_lazies()
+    out = torch.empty_like(local)
+    with pytest.raises(ValueError, match="checksum mismatch"):
+        weighted_combine_(out, [local, obj], [0.5, 0.5])
+    tensor_codec.release_parts(extras)
+
+
+@needs_gpu
+def test_lazy_ipc_materialize_fallback():
+    """3-input combines (or non-bf16) materialize the lazy and still agree
+    with the eager path."""
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+    from rayfed_amd.parallel.fedavg import weighted_combine_
+
+    plane = GpuDataPlane(GpuDataPlaneConfig(lazy_ipc=True))
+    a = torch.randn(1 << 20, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(1 << 20, dtype=torch.bfloat16, device="cuda")
+    c = torch.randn(1 << 20, dtype=torch.bfloat16, device="cuda")
+    extras, parts = tensor_codec.encode(b, plane, shm=True)
+    payload = b"".join(bytes(p) for p in parts)
+    lazy = tensor_codec.decode(
+        extras, memoryview(payload), plane, None, allow_lazy=True
+    )
+    plane.pop_pending_lazies()
+    out = torch.empty_like(a)
+    weighted_combine_(out, [a, lazy, c], [0.2, 0.3, 0.5])
+    ref = torch.empty_like(a)
+    weighted_combine_(ref, [a, b, c], [0.2, 0.3, 0.5])
+    assert torch.equal(out, ref)
+    tensor_codec.release_parts(extras)
