@@ -244,9 +244,12 @@ __device__ __forceinline__ void block_xor_out64(unsigned long long v,
   }
 }
 
+template <bool kPack>
 __global__ void hash64_kernel(const unsigned long long* __restrict__ words,
+                              unsigned long long* __restrict__ dst,
                               unsigned long long n_words,
                               const uint8_t* __restrict__ tail,
+                              uint8_t* __restrict__ dst_tail,
                               uint32_t tail_len,
                               unsigned long long nbytes,
                               unsigned long long* __restrict__ out) {
@@ -259,15 +262,28 @@ __global__ void hash64_kernel(const unsigned long long* __restrict__ words,
     h[k] = (kFnvOff ^ (unsigned long long)(lane * 4u + k)) * kFnvP;
   // Iteration j: lane reads words j*slots + lane*4 + k — consecutive lanes
   // read consecutive 32-byte chunks (fully coalesced); 4 independent
-  // accumulators hide the multiply latency.
+  // accumulators hide the multiply latency.  kPack fuses the staging copy
+  // into the same read pass (one HBM read instead of copy ∥ hash reading
+  // the source twice).
   for (unsigned long long base = (unsigned long long)lane * 4ull;
        base < n_words; base += slots) {
     if (base + 4 <= n_words) {
+      unsigned long long w[4];
 #pragma unroll
-      for (int k = 0; k < 4; ++k) h[k] = (h[k] ^ words[base + k]) * kFnvP;
+      for (int k = 0; k < 4; ++k) {
+        w[k] = words[base + k];
+        h[k] = (h[k] ^ w[k]) * kFnvP;
+      }
+      if (kPack) {
+#pragma unroll
+        for (int k = 0; k < 4; ++k) dst[base + k] = w[k];
+      }
     } else {
-      for (unsigned long long i = base; i < n_words; ++i)
-        h[i - base] = (h[i - base] ^ words[i]) * kFnvP;
+      for (unsigned long long i = base; i < n_words; ++i) {
+        const unsigned long long w = words[i];
+        h[i - base] = (h[i - base] ^ w) * kFnvP;
+        if (kPack) dst[i] = w;
+      }
     }
   }
   unsigned long long hl =
@@ -276,8 +292,10 @@ __global__ void hash64_kernel(const unsigned long long* __restrict__ words,
   if (lane == 0) {
     if (tail_len) {
       unsigned long long tw = 0;
-      for (uint32_t i = 0; i < tail_len; ++i)
+      for (uint32_t i = 0; i < tail_len; ++i) {
         tw |= (unsigned long long)tail[i] << (8 * i);
+        if (kPack) dst_tail[i] = tail[i];
+      }
       v ^= fmix64((kFnvOff ^ tw) * kFnvP);
     }
     v ^= fmix64((nbytes * kFnvP) ^ kFnvOff);
@@ -740,10 +758,38 @@ torch::Tensor hash64_async(torch::Tensor bytes) {
   const unsigned long long n_words = n / 8;
   const uint32_t tail_len = (uint32_t)(n % 8);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(hash64_kernel, dim3(kHashLanes / kBlock), dim3(kBlock),
-                     0, stream,
-                     reinterpret_cast<const unsigned long long*>(p), n_words,
-                     p + n_words * 8, tail_len, n,
+  hipLaunchKernelGGL(hash64_kernel<false>, dim3(kHashLanes / kBlock),
+                     dim3(kBlock), 0, stream,
+                     reinterpret_cast<const unsigned long long*>(p), nullptr,
+                     n_words, p + n_words * 8, nullptr, tail_len, n,
+                     reinterpret_cast<unsigned long long*>(
+                         out.data_ptr<int64_t>()));
+  return out;
+}
+
+torch::Tensor pack_hash64_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kUInt8 &&
+                  src.is_contiguous(),
+              "pack_hash64 expects contiguous CUDA uint8 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kUInt8 &&
+                  dst.is_contiguous() && dst.numel() >= src.numel(),
+              "pack_hash64 dst too small");
+  const unsigned long long n = src.numel();
+  auto out = torch::zeros(
+      {1}, torch::dtype(torch::kInt64).device(src.device()));
+  const uint8_t* p = src.data_ptr<uint8_t>();
+  uint8_t* q = dst.data_ptr<uint8_t>();
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(p) & 7u) == 0 &&
+                  (reinterpret_cast<uintptr_t>(q) & 7u) == 0,
+              "pack_hash64 requires 8-byte alignment");
+  const unsigned long long n_words = n / 8;
+  const uint32_t tail_len = (uint32_t)(n % 8);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(hash64_kernel<true>, dim3(kHashLanes / kBlock),
+                     dim3(kBlock), 0, stream,
+                     reinterpret_cast<const unsigned long long*>(p),
+                     reinterpret_cast<unsigned long long*>(q), n_words,
+                     p + n_words * 8, q + n_words * 8, tail_len, n,
                      reinterpret_cast<unsigned long long*>(
                          out.data_ptr<int64_t>()));
   return out;
@@ -1013,6 +1059,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "CRC32 of a device uint8 tensor -> int32[3] device tensor "
         "(out[2] = finalized CRC)");
   m.def("crc32", &crc32_sync, "CRC32 of a device uint8 tensor (synchronizes)");
+  m.def("pack_hash64_async", &pack_hash64_async,
+        "fused copy src->dst + hash64 of the bytes (one HBM read pass) -> "
+        "int64[1] device tensor");
   m.def("hash64_async", &hash64_async,
         "memory-rate 64-bit FNV/murmur integrity hash -> int64[1] device "
         "tensor (device-IPC lane checksum; numpy reference in "

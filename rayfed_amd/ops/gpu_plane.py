@@ -261,11 +261,25 @@ class GpuDataPlane:
                     )
             else:
                 flat = t.view(-1).view(torch.uint8)
+                src = flat[lo:hi]
+                if (
+                    self.config.verify_crc
+                    and self.config.device_checksum == "fnv64"
+                    and (src.data_ptr() & 7) == 0
+                ):
+                    # Fused copy+hash: ONE read of the source instead of
+                    # the copy and the checksum pass each reading it.
+                    with torch.cuda.stream(self._copy_stream):
+                        crc_outs.append((
+                            "fnv64",
+                            self._ext.pack_hash64_async(src, slab[3][: hi - lo]),
+                        ))
+                    continue
                 with torch.cuda.stream(self._copy_stream):
-                    slab[3][: hi - lo].copy_(flat[lo:hi])
+                    slab[3][: hi - lo].copy_(src)
                 if self.config.verify_crc:
                     with torch.cuda.stream(self._crc_stream):
-                        crc_outs.append(self._ck_async(flat[lo:hi]))
+                        crc_outs.append(self._ck_async(src))
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
@@ -423,6 +437,18 @@ class GpuDataPlane:
                     self._ext.unpack_fp8_async(src, out.view(-1)[lo:hi])
             else:
                 flat = out.view(-1).view(torch.uint8)
+                if (
+                    self.config.verify_crc
+                    and crcs is not None
+                    and cks[i] == "fnv64"
+                    and (flat[lo:hi].data_ptr() & 7) == 0
+                ):
+                    # Fused copy-out + verify hash: one read of the slab.
+                    with torch.cuda.stream(self._copy_stream):
+                        crc_outs.append(
+                            self._ext.pack_hash64_async(src, flat[lo:hi])
+                        )
+                    continue
                 with torch.cuda.stream(self._copy_stream):
                     flat[lo:hi].copy_(src)
             if self.config.verify_crc and crcs is not None:

@@ -595,3 +595,18 @@ def test_lazy_ipc_materialize_fallback():
     weighted_combine_(ref, [a, b, c], [0.2, 0.3, 0.5])
     assert torch.equal(out, ref)
     tensor_codec.release_parts(extras)
+
+
+@needs_gpu
+def test_pack_hash64_copies_and_hashes(ext):
+    from rayfed_amd.ops.hash_ref import hash64_ref
+
+    for n in (8, 4096, (1 << 20) + 24):
+        torch.manual_seed(n)
+        src = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+        dst = torch.zeros(n, dtype=torch.uint8, device="cuda")
+        hv = ext.pack_hash64_async(src, dst)
+        torch.cuda.synchronize()
+        assert torch.equal(dst, src)
+        got = int(hv.item()) & 0xFFFFFFFFFFFFFFFF
+        assert got == hash64_ref(src.cpu().numpy().tobytes())
