@@ -280,3 +280,20 @@ def test_chunk_streamed_send_roundtrip(monkeypatch):
     finally:
         send.stop()
         recv.stop()
+
+
+def test_gil_switch_interval_tuned_and_restored():
+    """fed.init lowers the GIL switch interval (hot-path wakeup latency);
+    shutdown restores it; RAYFED_GIL_SWITCH_US=0 opts out."""
+    import sys
+
+    import rayfed_amd as fed
+
+    base = sys.getswitchinterval()
+    fed.init(addresses=make_addresses(["alice"]), party="alice",
+             logging_level="warning")
+    try:
+        assert abs(sys.getswitchinterval() - 0.00025) < 1e-6
+    finally:
+        fed.shutdown()
+    assert abs(sys.getswitchinterval() - base) < 1e-9
