@@ -123,7 +123,7 @@ class PinnedPool {
   void release(std::shared_ptr<PinnedBuf> b) {
     std::lock_guard<std::mutex> lk(mu_);
     auto& bucket = free_[b->cap];
-    size_t limit = b->cap >= (64u << 20) ? 24 : 8;
+    size_t limit = b->cap >= (64u << 20) ? 16 : 8;  // <= ~4 GiB pinned held
     if (bucket.size() < limit) bucket.push_back(std::move(b));
     // else: shared_ptr drops it; ~PinnedBuf frees
   }
