@@ -347,8 +347,22 @@ class GpuDataPlane:
                 self._crc_stream.wait_event(produced)
         for f, (si, off) in zip(flats, placements):
             n = f.numel()
+            dstv = slabs[si][3][off : off + n]
+            if self.config.verify_crc and self.config.device_checksum == "fnv64":
+                # fnv64 arena path: fused copy+hash when the source is
+                # 8-byte aligned; otherwise copy then hash the (always
+                # aligned) slab region on the same stream — either way the
+                # LDS-bound CRC32 never runs here.
+                with torch.cuda.stream(self._copy_stream):
+                    if (f.data_ptr() & 7) == 0:
+                        out = self._ext.pack_hash64_async(f, dstv)
+                    else:
+                        dstv.copy_(f)
+                        out = self._ext.hash64_async(dstv)
+                crc_outs.append(("fnv64", out))
+                continue
             with torch.cuda.stream(self._copy_stream):
-                slabs[si][3][off : off + n].copy_(f)
+                dstv.copy_(f)
             if self.config.verify_crc:
                 with torch.cuda.stream(self._crc_stream):
                     crc_outs.append(self._ck_async(f))
@@ -388,16 +402,26 @@ class GpuDataPlane:
         ck = man.get("ck", "crc32")
         crc_out = None
         flat = out.view(-1).view(torch.uint8)
+        fuse = (
+            self.config.verify_crc
+            and crc_expect is not None
+            and ck == "fnv64"
+            and (src.data_ptr() & 7) == 0
+        )
         with torch.cuda.stream(self._copy_stream):
-            flat.copy_(src)
+            if fuse:
+                crc_out = self._ext.pack_hash64_async(src, flat)
+            else:
+                flat.copy_(src)
             done = self._copy_stream.record_event()
-        if self.config.verify_crc and crc_expect is not None:
+        if not fuse and self.config.verify_crc and crc_expect is not None:
             with torch.cuda.stream(self._crc_stream):
                 crc_out = self._ck_verify_async(ck, src)
                 crc_done = self._crc_stream.record_event()
         done.synchronize()
         if crc_out is not None:
-            crc_done.synchronize()
+            if not fuse:
+                crc_done.synchronize()
             got = self._ck_value(ck, crc_out)
             if got != crc_expect:
                 raise ValueError(
