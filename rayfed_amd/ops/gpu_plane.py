@@ -375,11 +375,21 @@ class GpuDataPlane:
             crc_done.synchronize()
         group = {"slabs": [s[1] for s in slabs], "slab_bytes": S}
         fields = []
+        vals = None
+        if self.config.verify_crc and crc_outs and all(
+            k == "fnv64" for k, _ in crc_outs
+        ):
+            # ONE batched D2H for the whole group's checksum values.
+            vals = torch.cat([o for _, o in crc_outs]).cpu().tolist()
         for i, (si, off) in enumerate(placements):
             f = {"slab": si, "off": off}
             if self.config.verify_crc:
                 kind, out = crc_outs[i]
-                f["crc32"] = self._ck_value(kind, out)
+                f["crc32"] = (
+                    int(vals[i]) & 0xFFFFFFFFFFFFFFFF
+                    if vals is not None
+                    else self._ck_value(kind, out)
+                )
                 f["ck"] = kind
             fields.append(f)
 
@@ -388,6 +398,62 @@ class GpuDataPlane:
                 self._ipc_put(s)
 
         return group, fields, release
+
+    def unpack_from_ipc_group_batch(self, group, mans, dtypes, shapes):
+        """Arena receive of MANY tensors with one sync and one batched
+        checksum read-back — a 291-tensor state dict costs 2 host round
+        trips instead of 582 (per-tensor sync + per-value .item())."""
+        self._bind_device()
+        outs = []
+        ck_outs = []  # (i, kind, out_tensor)
+        crc_pairs = []  # for crc32 entries on the crc stream
+        for i, (man, dtype, shape) in enumerate(zip(mans, dtypes, shapes)):
+            nbytes = man["nbytes"]
+            handle = bytes(group["slabs"][man["slab"]])
+            off = man["off"]
+            src = self._ipc_src_view(handle, off + nbytes)[off : off + nbytes]
+            out = torch.empty(shape, dtype=dtype, device=self.device)
+            outs.append(out)
+            flat = out.view(-1).view(torch.uint8)
+            expect = man.get("crc32")
+            ck = man.get("ck", "crc32")
+            verify = self.config.verify_crc and expect is not None
+            with torch.cuda.stream(self._copy_stream):
+                if verify and ck == "fnv64" and (src.data_ptr() & 7) == 0:
+                    ck_outs.append(
+                        (i, "fnv64", self._ext.pack_hash64_async(src, flat))
+                    )
+                    continue
+                flat.copy_(src)
+            if verify:
+                with torch.cuda.stream(self._crc_stream):
+                    ck_outs.append((i, ck, self._ck_verify_async(ck, src)))
+                    crc_pairs.append(i)
+        with torch.cuda.stream(self._copy_stream):
+            done = self._copy_stream.record_event()
+        if crc_pairs:
+            with torch.cuda.stream(self._crc_stream):
+                crc_done = self._crc_stream.record_event()
+        done.synchronize()
+        if crc_pairs:
+            crc_done.synchronize()
+        if ck_outs:
+            if all(o[1] == "fnv64" for o in ck_outs):
+                # ONE batched D2H for every checksum value.
+                vals = torch.cat([o[2] for o in ck_outs]).cpu().tolist()
+            else:  # mixed/crc32 — rare (misaligned sources only)
+                vals = [self._ck_value(o[1], o[2]) for o in ck_outs]
+            for (i, kind, _raw), v in zip(ck_outs, vals):
+                got = int(v) & (
+                    0xFFFFFFFFFFFFFFFF if kind == "fnv64" else 0xFFFFFFFF
+                )
+                expect = mans[i].get("crc32")
+                if got != expect:
+                    raise ValueError(
+                        f"GPU tensor checksum mismatch (ipc group, tensor "
+                        f"{i}): expected {expect:#x}, got {got:#x}"
+                    )
+        return outs
 
     def unpack_from_ipc_group(self, group, man, dtype, shape):
         """Receiver of an arena-packed tensor: D2D from (slab, off) with the

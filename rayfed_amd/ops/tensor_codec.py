@@ -452,20 +452,30 @@ def decode(
     skeleton = payload[:skel_len]
     off = skel_len
     tensors: List[Any] = []
-    for man in extras["tensors"]:
+    # Arena-grouped tensors decode as ONE batch (one sync + one checksum
+    # read-back for the whole group — a 291-tensor state dict would
+    # otherwise pay a host round trip per tensor).
+    group_batch: Dict[int, Any] = {}
+    ipcg_idx = [
+        i for i, m in enumerate(extras["tensors"]) if m.get("ipcg")
+    ]
+    if ipcg_idx:
+        if gpu_plane is None:
+            raise RuntimeError(
+                "received a device-IPC tensor but no GPU data plane is "
+                "attached (set RAYFED_IPC=0 on the sender for CPU peers)"
+            )
+        mans = [extras["tensors"][i] for i in ipcg_idx]
+        outs = gpu_plane.unpack_from_ipc_group_batch(
+            extras["ipc_group"], mans,
+            [_STR_TO_DTYPE[m["dtype"]] for m in mans],
+            [m["shape"] for m in mans],
+        )
+        group_batch = dict(zip(ipcg_idx, outs))
+    for idx, man in enumerate(extras["tensors"]):
         nbytes = man["nbytes"]
         if man.get("ipcg"):
-            if gpu_plane is None:
-                raise RuntimeError(
-                    "received a device-IPC tensor but no GPU data plane is "
-                    "attached (set RAYFED_IPC=0 on the sender for CPU peers)"
-                )
-            tensors.append(
-                gpu_plane.unpack_from_ipc_group(
-                    extras["ipc_group"], man, _STR_TO_DTYPE[man["dtype"]],
-                    man["shape"],
-                )
-            )
+            tensors.append(group_batch[idx])
             continue
         if "ipc_slabs" in man:
             if gpu_plane is None:
