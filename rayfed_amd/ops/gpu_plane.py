@@ -1019,6 +1019,12 @@ class LazyIpcTensor:
         except Exception:  # noqa: BLE001
             pass
 
+    def __reduce__(self):
+        raise TypeError(
+            "LazyIpcTensor cannot be re-serialized (it borrows the sender's "
+            "IPC slabs) — materialize() it before sending it onward"
+        )
+
     def __repr__(self):
         return (f"LazyIpcTensor(shape={self.shape}, dtype={self.dtype}, "
                 f"released={self._released})")
