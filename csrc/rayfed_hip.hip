@@ -479,6 +479,117 @@ __global__ void pack_fp8_kernel(const uint16_t* __restrict__ src,
   block_xor_out(contrib_tail, &out[1], lds_scratch);
 }
 
+// fp8 wire cast fused with the hash64 slot mapping over the PRODUCED fp8
+// bytes (the LDS-table CRC in pack_fp8_kernel is the bottleneck otherwise).
+// Output word k packs 8 fp8 from 8 bf16 (two input u64 words).
+__global__ void pack_fp8_hash64_kernel(const uint16_t* __restrict__ src,
+                                       unsigned long long* __restrict__ dst,
+                                       unsigned long long n_words,  // output
+                                       uint32_t tail_elems,
+                                       unsigned long long nbytes,   // output
+                                       unsigned long long* __restrict__ out) {
+  __shared__ unsigned long long lds[kBlock / kWave];
+  const uint32_t lane = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long slots = 4ull * kHashLanes;
+  unsigned long long h[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k)
+    h[k] = (kFnvOff ^ (unsigned long long)(lane * 4u + k)) * kFnvP;
+  for (unsigned long long base = (unsigned long long)lane * 4ull;
+       base < n_words; base += slots) {
+    const int cnt = (int)((base + 4 <= n_words) ? 4 : (n_words - base));
+    for (int k = 0; k < cnt; ++k) {
+      const unsigned long long i = base + k;
+      const uint4 v = *reinterpret_cast<const uint4*>(src + i * 8);  // 8 bf16
+      unsigned long long w = 0;
+      const uint32_t parts[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        w |= (unsigned long long)f32_to_fp8(
+                 bf16_to_f32((uint16_t)(parts[e] & 0xFFFFu)))
+             << (16 * e);
+        w |= (unsigned long long)f32_to_fp8(
+                 bf16_to_f32((uint16_t)(parts[e] >> 16)))
+             << (16 * e + 8);
+      }
+      h[k] = (h[k] ^ w) * kFnvP;
+      dst[i] = w;
+    }
+  }
+  unsigned long long hl =
+      ((((h[0] * kFnvP ^ h[1]) * kFnvP ^ h[2]) * kFnvP ^ h[3]) * kFnvP);
+  unsigned long long v = fmix64(hl);
+  if (lane == 0) {
+    if (tail_elems) {
+      unsigned long long tw = 0;
+      uint8_t* dtail = reinterpret_cast<uint8_t*>(dst + n_words);
+      for (uint32_t e = 0; e < tail_elems; ++e) {
+        const uint8_t b = f32_to_fp8(bf16_to_f32(src[n_words * 8 + e]));
+        dtail[e] = b;
+        tw |= (unsigned long long)b << (8 * e);
+      }
+      v ^= fmix64((kFnvOff ^ tw) * kFnvP);
+    }
+    v ^= fmix64((nbytes * kFnvP) ^ kFnvOff);
+  }
+  block_xor_out64(v, out, lds);
+}
+
+// Inverse: fp8 -> bf16 expand fused with the hash64 of the INPUT fp8 bytes
+// (receiver-side verify without a second read).
+__global__ void unpack_fp8_hash64_kernel(
+    const unsigned long long* __restrict__ src, uint16_t* __restrict__ dst,
+    unsigned long long n_words,  // input u64 words
+    uint32_t tail_elems, unsigned long long nbytes,
+    unsigned long long* __restrict__ out) {
+  __shared__ unsigned long long lds[kBlock / kWave];
+  const uint32_t lane = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned long long slots = 4ull * kHashLanes;
+  unsigned long long h[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k)
+    h[k] = (kFnvOff ^ (unsigned long long)(lane * 4u + k)) * kFnvP;
+  for (unsigned long long base = (unsigned long long)lane * 4ull;
+       base < n_words; base += slots) {
+    const int cnt = (int)((base + 4 <= n_words) ? 4 : (n_words - base));
+    for (int k = 0; k < cnt; ++k) {
+      const unsigned long long i = base + k;
+      const unsigned long long w = src[i];
+      h[k] = (h[k] ^ w) * kFnvP;
+      uint4 o;
+      uint32_t parts[4];
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        parts[e] =
+            (uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(w >> (16 * e)))) |
+            ((uint32_t)f32_to_bf16(fp8_to_f32((uint8_t)(w >> (16 * e + 8))))
+             << 16);
+      }
+      o.x = parts[0];
+      o.y = parts[1];
+      o.z = parts[2];
+      o.w = parts[3];
+      *reinterpret_cast<uint4*>(dst + i * 8) = o;
+    }
+  }
+  unsigned long long hl =
+      ((((h[0] * kFnvP ^ h[1]) * kFnvP ^ h[2]) * kFnvP ^ h[3]) * kFnvP);
+  unsigned long long v = fmix64(hl);
+  if (lane == 0) {
+    if (tail_elems) {
+      unsigned long long tw = 0;
+      const uint8_t* stail = reinterpret_cast<const uint8_t*>(src + n_words);
+      for (uint32_t e = 0; e < tail_elems; ++e) {
+        tw |= (unsigned long long)stail[e] << (8 * e);
+        dst[n_words * 8 + e] = f32_to_bf16(fp8_to_f32(stail[e]));
+      }
+      v ^= fmix64((kFnvOff ^ tw) * kFnvP);
+    }
+    v ^= fmix64((nbytes * kFnvP) ^ kFnvOff);
+  }
+  block_xor_out64(v, out, lds);
+}
+
 __global__ void unpack_fp8_kernel(const uint8_t* __restrict__ src,
                                   uint16_t* __restrict__ dst,
                                   unsigned long long n_elems) {
@@ -851,6 +962,60 @@ torch::Tensor pack_fp8_async(torch::Tensor src, torch::Tensor dst) {
   return out;
 }
 
+torch::Tensor pack_fp8_hash64_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kBFloat16 &&
+                  src.is_contiguous(),
+              "pack_fp8_hash64 expects contiguous CUDA bf16 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kUInt8 &&
+                  dst.is_contiguous() && dst.numel() >= src.numel(),
+              "pack_fp8_hash64 dst too small");
+  const unsigned long long n = src.numel();  // fp8 output bytes
+  auto out = torch::zeros(
+      {1}, torch::dtype(torch::kInt64).device(src.device()));
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(src.data_ptr()) & 15u) == 0 &&
+                  (reinterpret_cast<uintptr_t>(dst.data_ptr()) & 7u) == 0,
+              "pack_fp8_hash64 alignment");
+  const unsigned long long n_words = n / 8;
+  const uint32_t tail = (uint32_t)(n % 8);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(pack_fp8_hash64_kernel, dim3(kHashLanes / kBlock),
+                     dim3(kBlock), 0, stream,
+                     reinterpret_cast<const uint16_t*>(src.data_ptr()),
+                     reinterpret_cast<unsigned long long*>(
+                         dst.data_ptr<uint8_t>()),
+                     n_words, tail, n,
+                     reinterpret_cast<unsigned long long*>(
+                         out.data_ptr<int64_t>()));
+  return out;
+}
+
+torch::Tensor unpack_fp8_hash64_async(torch::Tensor src, torch::Tensor dst) {
+  TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kUInt8 &&
+                  src.is_contiguous(),
+              "unpack_fp8_hash64 expects contiguous CUDA uint8 src");
+  TORCH_CHECK(dst.is_cuda() && dst.dtype() == torch::kBFloat16 &&
+                  dst.is_contiguous() && dst.numel() == src.numel(),
+              "unpack_fp8_hash64 dst mismatch");
+  const unsigned long long n = src.numel();
+  auto out = torch::zeros(
+      {1}, torch::dtype(torch::kInt64).device(src.device()));
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(src.data_ptr()) & 7u) == 0 &&
+                  (reinterpret_cast<uintptr_t>(dst.data_ptr()) & 15u) == 0,
+              "unpack_fp8_hash64 alignment");
+  const unsigned long long n_words = n / 8;
+  const uint32_t tail = (uint32_t)(n % 8);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(unpack_fp8_hash64_kernel, dim3(kHashLanes / kBlock),
+                     dim3(kBlock), 0, stream,
+                     reinterpret_cast<const unsigned long long*>(
+                         src.data_ptr<uint8_t>()),
+                     reinterpret_cast<uint16_t*>(dst.data_ptr()), n_words,
+                     tail, n,
+                     reinterpret_cast<unsigned long long*>(
+                         out.data_ptr<int64_t>()));
+  return out;
+}
+
 void unpack_fp8_async(torch::Tensor src, torch::Tensor dst) {
   TORCH_CHECK(src.is_cuda() && src.dtype() == torch::kUInt8 &&
                   src.is_contiguous(),
@@ -1071,6 +1236,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_fp8_async", &pack_fp8_async,
         "fused bf16 -> OCP fp8 e4m3 cast + CRC32 of the fp8 bytes");
   m.def("unpack_fp8_async", &unpack_fp8_async, "fp8 e4m3 bytes -> bf16");
+  m.def("pack_fp8_hash64_async", &pack_fp8_hash64_async,
+        "bf16 -> fp8 e4m3 cast fused with hash64 of the fp8 bytes");
+  m.def("unpack_fp8_hash64_async", &unpack_fp8_hash64_async,
+        "fp8 e4m3 bytes -> bf16 fused with hash64 of the fp8 bytes");
   m.def("fedavg_reduce_", &fedavg_reduce_,
         "out = sum_k w_k * in_k (bf16/f16/f32, fp32 accumulation)");
   m.def("fedavg_combine_hash_async", &fedavg_combine_hash_async,

@@ -252,13 +252,24 @@ class GpuDataPlane:
         for i, slab in enumerate(slabs):
             lo, hi = i * S, min((i + 1) * S, nbytes)
             if wire_fp8:
-                # fp8 cast must produce the wire bytes anyway; keep fused.
+                # fp8 cast produces the wire bytes anyway; fuse the checksum
+                # of those bytes into the same pass (hash64 preferred — the
+                # LDS-table CRC dominates the cast kernel otherwise).
                 with torch.cuda.stream(self._copy_stream):
-                    crc_outs.append(
-                        self._ext.pack_fp8_async(
-                            t.view(-1)[lo:hi], slab[3][: hi - lo]
-                        )
-                    )
+                    if self.config.device_checksum == "fnv64":
+                        crc_outs.append((
+                            "fnv64",
+                            self._ext.pack_fp8_hash64_async(
+                                t.view(-1)[lo:hi], slab[3][: hi - lo]
+                            ),
+                        ))
+                    else:
+                        crc_outs.append((
+                            "crc32",
+                            self._ext.pack_fp8_async(
+                                t.view(-1)[lo:hi], slab[3][: hi - lo]
+                            ),
+                        ))
             else:
                 flat = t.view(-1).view(torch.uint8)
                 src = flat[lo:hi]
@@ -289,12 +300,8 @@ class GpuDataPlane:
             crc_done.synchronize()
         crcs = cks = None
         if self.config.verify_crc and crc_outs:
-            if wire_fp8:  # pack_fp8 computes CRC32 inline
-                crcs = [int(c[2].item()) & 0xFFFFFFFF for c in crc_outs]
-                cks = ["crc32"] * len(crc_outs)
-            else:
-                crcs = [self._ck_value(k, o) for k, o in crc_outs]
-                cks = [k for k, _ in crc_outs]
+            crcs = [self._ck_value(k, o) for k, o in crc_outs]
+            cks = [k for k, _ in crc_outs]
         man = {"ipc_slabs": [s[1] for s in slabs], "slab_bytes": S}
         if crcs is not None:
             man["ipc_crcs"] = crcs
@@ -524,6 +531,16 @@ class GpuDataPlane:
             src = self._ipc_src_view(h, hi - lo)[: hi - lo]
             if wire_fp8:
                 with torch.cuda.stream(self._copy_stream):
+                    if (
+                        self.config.verify_crc
+                        and crcs is not None
+                        and cks[i] == "fnv64"
+                    ):
+                        # Fused expand + verify hash of the fp8 bytes.
+                        crc_outs.append(self._ext.unpack_fp8_hash64_async(
+                            src, out.view(-1)[lo:hi]
+                        ))
+                        continue
                     self._ext.unpack_fp8_async(src, out.view(-1)[lo:hi])
             else:
                 flat = out.view(-1).view(torch.uint8)

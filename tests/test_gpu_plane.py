@@ -692,3 +692,30 @@ def test_concurrent_multi_gib_pushes(monkeypatch):
     finally:
         send.stop()
         recv.stop()
+
+
+@needs_gpu
+def test_fp8_hash64_fused_kernels(ext):
+    """Fused fp8 cast+hash64 and expand+hash64: fp8 bytes identical to the
+    CRC-era pack_fp8 kernel, hash matches the numpy reference over those
+    bytes, expand round-trips."""
+    from rayfed_amd.ops.hash_ref import hash64_ref
+
+    for n in (64, 4096, (1 << 20) + 3):
+        torch.manual_seed(n)
+        src = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+        f8_a = torch.empty(n, dtype=torch.uint8, device="cuda")
+        f8_b = torch.empty(n, dtype=torch.uint8, device="cuda")
+        ext.pack_fp8_async(src, f8_a)
+        hv = ext.pack_fp8_hash64_async(src, f8_b)
+        torch.cuda.synchronize()
+        assert torch.equal(f8_a, f8_b), f"n={n}: fp8 bytes differ"
+        expect = hash64_ref(f8_b.cpu().numpy().tobytes())
+        assert (int(hv.item()) & 0xFFFFFFFFFFFFFFFF) == expect
+        back = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+        hv2 = ext.unpack_fp8_hash64_async(f8_b, back)
+        ref = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+        ext.unpack_fp8_async(f8_b, ref)
+        torch.cuda.synchronize()
+        assert torch.equal(back, ref)
+        assert (int(hv2.item()) & 0xFFFFFFFFFFFFFFFF) == expect
