@@ -536,16 +536,23 @@ def _run_extras() -> dict:
 
     out = {}
     specs = [
-        ("push", ["--mode", "push", "--steps", "8", "--warmup", "2"], 360),
-        ("fedavg", ["--mode", "fedavg", "--steps", "5", "--warmup", "1"], 360),
+        ("push", ["--mode", "push", "--steps", "8", "--warmup", "2"], 360, {}),
+        ("fedavg", ["--mode", "fedavg", "--steps", "5", "--warmup", "1"], 360,
+         {}),
+        # Cross-host-shaped lane: same push with the same-host fast lanes
+        # disabled, so the striped/chunk-streamed socket path is measured.
+        ("socket_push",
+         ["--mode", "push", "--steps", "4", "--warmup", "1",
+          "--push-gib", "2"], 360, {"RAYFED_SHM": "0"}),
     ]
     env = dict(os.environ)
     env["RAYFED_BENCH_EXTRAS"] = "0"
-    for name, flags, tmo in specs:
+    for name, flags, tmo, extra_env in specs:
         try:
             r = subprocess.run(
                 [sys.executable, os.path.abspath(__file__)] + flags,
-                capture_output=True, text=True, timeout=tmo, env=env,
+                capture_output=True, text=True, timeout=tmo,
+                env={**env, **extra_env},
                 cwd=os.path.dirname(os.path.abspath(__file__)),
             )
             line = next(

@@ -215,7 +215,9 @@ __device__ __forceinline__ uint16_t f32_to_bf16(float f) {
 // agree for any launch, and tests pin it against a numpy reference
 // (rayfed_amd/ops/hash_ref.py).  Error-detection only — NOT crypto.
 // ---------------------------------------------------------------------------
-constexpr uint32_t kHashLanes = 262144;  // fixed: 1024 blocks x 256 threads
+constexpr uint32_t kHashLanes = 524288;  // fixed: 2048 blocks x 256 threads
+// (8 waves/CU — 4 was latency-limited; the mapping constant is mirrored in
+// rayfed_amd/ops/hash_ref.py and must change in lockstep)
 constexpr unsigned long long kFnvOff = 0xcbf29ce484222325ull;
 constexpr unsigned long long kFnvP = 0x100000001b3ull;
 

@@ -1,6 +1,6 @@
 """CPU reference of the device hash64 kernel (csrc/rayfed_hip.hip).
 
-The GPU kernel is FNV-1a over SLOTS = 4 * 262144 interleaved u64 word
+The GPU kernel is FNV-1a over SLOTS = 4 * 524288 interleaved u64 word
 streams (slot s owns words {j*SLOTS + s}), a murmur fmix64 finalizer per
 lane, an XOR combine across lanes, plus tail-bytes and length terms.  The
 mapping depends only on nbytes, so this numpy implementation must produce
@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import numpy as np
 
-LANES = 262144
+LANES = 524288  # must equal kHashLanes in csrc/rayfed_hip.hip
 SLOTS = 4 * LANES
 FNV_OFF = np.uint64(0xCBF29CE484222325)
 FNV_P = np.uint64(0x100000001B3)
