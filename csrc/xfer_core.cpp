@@ -93,6 +93,14 @@ class PinnedPool {
     return pool;
   }
 
+  // Static-destruction order vs the HIP runtime is undefined at process
+  // exit — deliberately leak pooled buffers instead of calling
+  // hipHostFree after the runtime may have unloaded.
+  ~PinnedPool() {
+    for (auto& [cls, bucket] : free_)
+      for (auto& b : bucket) b->p = nullptr;
+  }
+
   std::shared_ptr<PinnedBuf> acquire(size_t n) {
     size_t cls = 1ull << (64 - __builtin_clzll(std::max<size_t>(n, 1 << 16) - 1));
     {
@@ -912,6 +920,7 @@ class XferClient {
                  uint64_t total, double timeout_s, std::string* result_out) {
     std::string key = host + ":" + std::to_string(port);
     std::unique_ptr<Stream> st = acquire_tls_conn(key, host, port);
+    set_recv_timeout(st->fd, std::max(1.0, timeout_s));
     uint64_t req_id = tls_req_id_.fetch_add(1);
     try {
       char head[16];
@@ -1139,11 +1148,21 @@ class XferClient {
     return {pending->code, pending->result};
   }
 
+  static void set_recv_timeout(int fd, double timeout_s) {
+    timeval tv;
+    tv.tv_sec = (time_t)timeout_s;
+    tv.tv_usec = (suseconds_t)((timeout_s - (double)tv.tv_sec) * 1e6);
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  }
+
   int send_tls(const std::string& host, int port, const std::string& preamble,
                const std::vector<std::pair<const char*, size_t>>& views,
                uint64_t total, double timeout_s, std::string* result_out) {
     std::string key = host + ":" + std::to_string(port);
     std::unique_ptr<Stream> st = acquire_tls_conn(key, host, port);
+    // Bound the ack wait: a half-open peer must surface as a broken
+    // connection (retryable) instead of hanging the exchange forever.
+    set_recv_timeout(st->fd, std::max(1.0, timeout_s));
     uint64_t req_id = tls_req_id_.fetch_add(1);
     try {
       char head[16];

@@ -200,11 +200,7 @@ class XferSenderService:
         up = str(upstream_seq_id)
         down = str(downstream_seq_id)
         if self._tls:
-            # send_async is plaintext-only (an SSL* cannot take a concurrent
-            # reader+writer); TLS keeps the pooled blocking path.
-            return self._pool.submit(
-                self._send_blocking, dest_party, data, up, down
-            )
+            return self._send_tls_continuation(dest_party, data, up, down)
 
         # Continuation-style send: issue the frame the moment the payload is
         # ready, IN the producing thread (already hot) — a pool thread woken
@@ -294,6 +290,71 @@ class XferSenderService:
                 with self._stats_lock:
                     edge = self._edges.setdefault(dest_party, self._edge_cls())
                     edge.record(state["nbytes"], secs, err)
+
+        if isinstance(data, ObjectRef) and not data.future.done():
+            data.future.add_done_callback(_issue)
+        else:
+            _issue()
+        return _LazyFuture(_fetch)
+
+    def _send_tls_continuation(self, dest_party, data, up, down) -> Future:
+        """TLS lane: no pipelined send_async (an SSL* is exclusive), so a
+        small frame's whole write+ack exchange runs inline in the producing
+        thread — one quick attempt, pool fallback with the full retry
+        budget.  Saves the pool wake-from-idle like the plaintext path."""
+        state: dict = {}
+        ready = threading.Event()
+
+        def _issue(_f=None):
+            try:
+                d = data
+                if isinstance(d, ObjectRef):
+                    d = d.result()
+                body_parts, extras, defer_ack, nbytes = self._encode_frame(
+                    dest_party, d, up, down
+                )
+                if not defer_ack and nbytes <= self._INLINE_MAX:
+                    t0 = time.perf_counter()
+                    host, port = self._addresses[dest_party].rsplit(":", 1)
+                    try:
+                        code, result = self._client_ctl.send(
+                            host, int(port), up, down, body_parts, False, 2.0
+                        )
+                        tensor_codec.release_parts(extras)
+                        state["inline"] = (code, result, nbytes, t0)
+                        return
+                    except RuntimeError:
+                        pass  # cold/broken conn: pooled path retries
+                state["fut"] = self._pool.submit(
+                    self._send_parts, dest_party, body_parts, extras,
+                    defer_ack, nbytes, up, down,
+                )
+            except BaseException as e:  # noqa: BLE001
+                state["exc"] = e
+            finally:
+                ready.set()
+
+        def _fetch(timeout=None):
+            if not ready.wait(timeout):
+                raise FutureTimeoutError()
+            if "exc" in state:
+                raise state["exc"]
+            if "fut" in state:
+                return state["fut"].result(timeout)
+            code, result, nbytes, t0 = state["inline"]
+            err = not (200 <= code < 400)
+            with self._stats_lock:
+                edge = self._edges.setdefault(dest_party, self._edge_cls())
+                edge.record(nbytes, time.perf_counter() - t0, err)
+            if 400 <= code < 500:
+                raise RuntimeError(
+                    f"[{code}] send to {dest_party} rejected: {result}"
+                )
+            if code >= 500:
+                raise RuntimeError(
+                    f"[{code}] send to {dest_party} failed: {result}"
+                )
+            return True
 
         if isinstance(data, ObjectRef) and not data.future.done():
             data.future.add_done_callback(_issue)
