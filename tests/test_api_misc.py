@@ -114,3 +114,24 @@ def _driver_local_get_of_plain_ref(party, addresses):
 
 def test_get_plain_object_ref():
     run_parties(_driver_local_get_of_plain_ref, parties=("alice",))
+
+
+def test_examples_demo_runs():
+    """The README-demo example (examples/demo.py) runs end to end — the
+    reference's README.md:124-168 flow on this engine."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = os.path.join(repo, "examples", "demo.py")
+    pb = subprocess.Popen([sys.executable, script, "bob"],
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                          text=True, cwd=repo)
+    pa = subprocess.run([sys.executable, script, "alice"],
+                        capture_output=True, text=True, timeout=120, cwd=repo)
+    out_b, _ = pb.communicate(timeout=120)
+    assert pa.returncode == 0, pa.stdout + pa.stderr
+    assert pb.returncode == 0, out_b
+    assert "The result in party alice is 5" in pa.stdout
+    assert "The result in party bob is 5" in out_b
