@@ -392,7 +392,6 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
     import torch
     import torch.distributed as dist
 
-    from rayfed_amd.parallel.fedavg import BucketedAllReducer
 
     dist.init_process_group(backend="gloo", rank=rank, world_size=world)
     half = world // 2

@@ -1,20 +1,22 @@
-"""GPU data plane: device tensor ⇄ host staging for the cross-silo path.
+"""GPU data plane: device tensor ⇄ staging/slabs for the cross-silo path.
 
 MI355X-native replacement for the reference's CPU pickle of payloads
-(/root/reference/fed/proxy/grpc/grpc_proxy.py:202).  Pipeline per tensor
+(/root/reference/fed/proxy/grpc/grpc_proxy.py:202).  Lanes, fastest first
 (SURVEY.md §2.3 / §7 step 3):
 
-  send:  HIP CRC32 kernel on a dedicated stream (one HBM read pass,
-         parallel slice CRC + GF(2) combine — csrc/rayfed_hip.hip)
-         ∥ hipMemcpyAsync D2H into pinned staging on the copy stream
-         → bytes handed to the transport frame
-  recv:  frame bytes → pinned staging → hipMemcpyAsync H2D on a side
-         stream → HIP CRC32 verify on the device tensor
+  device-IPC (same node): fused copy+hash64 pass into pooled 1 GiB hipIpc
+         slabs; the receiver D2D-copies with the verify fused into the same
+         read — or, with ``lazy_ipc``, combines DIRECTLY from the sender's
+         slabs (fedavg_combine_hash kernel, no materialized copy)
+  /dev/shm (same node, CPU peers / fallback): hipHostRegister-ed pooled
+         segments, chunk-pipelined with a progress page
+  socket/host: checksum kernel ∥ D2H into pinned staging → transport frame;
+         receiver H2Ds with the device verify
 
-Optional wire compression (``wire_dtype='fp8e4m3'``): bf16 tensors are cast
-to OCP fp8 e4m3 by the fused pack_fp8 kernel (cast + CRC in one pass) and
-expanded back to bf16 on the receiver — half the bytes on the wire, opt-in
-because it is lossy.
+Checksums: hash64 (memory-rate 64-bit hash, csrc hash64_kernel; numpy
+reference in ops/hash_ref.py) on the device lanes; zlib CRC32 on host/wire
+paths and by config.  Optional lossy wire compression
+(``wire_dtype='fp8e4m3'``) casts bf16→fp8 fused with the wire checksum.
 
 The HIP extension is REQUIRED on a GPU box — this module raises instead of
 falling back to an eager path, so a GPU test can never silently pass on CPU
