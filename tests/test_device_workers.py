@@ -131,3 +131,45 @@ def test_workers_can_read_job_config():
         assert vals == ["alice", "alice"]
     finally:
         fed.shutdown()
+
+
+def _gpu_make(n):
+    import torch
+
+    return torch.arange(n, dtype=torch.float32, device="cuda")
+
+
+def _gpu_double(t):
+    return t * 2
+
+
+def _gpu_sum(t):
+    return float(t.sum())
+
+
+def test_worker_pool_gpu_device_resident():
+    """One worker pinned to GPU 0: keep=True results stay device-resident
+    in the worker's object table; chained tasks consume them in place."""
+    import pytest as _pytest
+
+    torch = _pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        _pytest.skip("requires MI355X")
+    from rayfed_amd.runtime.worker import DeviceWorkerPool, RemoteHandle
+
+    pool = DeviceWorkerPool(devices=[0])
+    try:
+        h = pool.submit(0, _gpu_make, (1024,), keep=True).result(timeout=120)
+        assert isinstance(h, RemoteHandle)
+        h2 = pool.submit(0, _gpu_double, (h,), keep=True).result(timeout=60)
+        total = pool.submit(0, _gpu_sum, (h2,)).result(timeout=60)
+        assert total == float(2 * sum(range(1024)))
+        t = pool.fetch(h2).result(timeout=60)
+        assert t.is_cuda and float(t[3]) == 6.0
+    finally:
+        pool.shutdown()
+
+
+test_worker_pool_gpu_device_resident = __import__("pytest").mark.gpu(
+    test_worker_pool_gpu_device_resident
+)
