@@ -147,11 +147,13 @@ def _gpu_sum(t):
     return float(t.sum())
 
 
+import pytest as _pytest
+
+
+@_pytest.mark.gpu
 def test_worker_pool_gpu_device_resident():
     """One worker pinned to GPU 0: keep=True results stay device-resident
     in the worker's object table; chained tasks consume them in place."""
-    import pytest as _pytest
-
     torch = _pytest.importorskip("torch")
     if not torch.cuda.is_available():
         _pytest.skip("requires MI355X")
@@ -168,8 +170,3 @@ def test_worker_pool_gpu_device_resident():
         assert t.is_cuda and float(t[3]) == 6.0
     finally:
         pool.shutdown()
-
-
-test_worker_pool_gpu_device_resident = __import__("pytest").mark.gpu(
-    test_worker_pool_gpu_device_resident
-)
