@@ -15,7 +15,7 @@ from torch.utils import cpp_extension  # noqa: E402
 
 setup(
     name="rayfed_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=find_packages(include=["rayfed_amd", "rayfed_amd.*"]),
     ext_modules=[
         cpp_extension.CUDAExtension(
