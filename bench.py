@@ -317,15 +317,32 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
     # Flat gradient buffer with per-parameter VIEWS (the flat-grads layout
     # production DDP keeps): the cross-party push and the intra-party
     # all-reduce both operate on `flat` directly — no per-round repack.
+    # RAYFED_BENCH_ARENA=1: the gradients LIVE in IPC-shared arena slabs
+    # instead — the exchange ships (handle, hash) per shard, zero pack.
     shapes = llama3_8b_grad_shapes(layers, vocab)
     total = sum(int(torch.prod(torch.tensor(sh))) for sh in shapes)
-    flat = torch.empty(total, dtype=dtype, device=dev).uniform_(-1, 1)
-    grads, off = [], 0
-    for sh in shapes:
-        n_el = int(torch.prod(torch.tensor(sh)))
-        grads.append(flat[off : off + n_el].view(sh))
-        off += n_el
-    nbytes = total * flat.element_size()
+    use_arena = os.environ.get("RAYFED_BENCH_ARENA") == "1" and use_gpu
+    if use_arena:
+        from rayfed_amd.ops.gpu_plane import maybe_create_gpu_plane
+
+        plane = maybe_create_gpu_plane({"lazy_ipc": True})
+        arena = plane.alloc_shared_arena(total * 2)
+        S = arena.SLAB // 2  # bf16 elements per slab
+        shards = []
+        left = total
+        for sv in arena.shards:
+            n_el = min(left, sv.numel() // 2)
+            if n_el <= 0:
+                break
+            shards.append(sv[: n_el * 2].view(dtype))
+            left -= n_el
+        flat = None
+        for sh in shards:
+            sh.uniform_(-1, 1)
+    else:
+        flat = torch.empty(total, dtype=dtype, device=dev).uniform_(-1, 1)
+        shards = None
+    nbytes = total * 2
     from rayfed_amd.parallel.fedavg import allreduce_flat_
 
     @fed.remote
@@ -333,6 +350,14 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
         """Receives the peer's averaged grads and combines with local."""
 
         def combine(self, peer_flat, local_flat):
+            if isinstance(local_flat, (list, tuple)):  # arena shards
+                acc = 0.0
+                for pv, lv in zip(peer_flat, local_flat):
+                    out = torch.empty_like(lv)
+                    weighted_combine_(out, [lv, pv], [0.5, 0.5])
+                    acc += float(out[:2].float().sum())
+                torch.cuda.synchronize()
+                return acc
             out = torch.empty_like(local_flat)
             weighted_combine_(out, [local_flat, peer_flat], [0.5, 0.5])
             if out.is_cuda:
@@ -349,8 +374,12 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
     def produce(_tick):
         if party_group is not None:
             # Intra-party RCCL over xGMI, in place on the flat buffer.
-            allreduce_flat_(flat, group=party_group)
-        return flat
+            if shards is not None:
+                for sh in shards:
+                    allreduce_flat_(sh, group=party_group)
+            else:
+                allreduce_flat_(flat, group=party_group)
+        return shards if shards is not None else flat
 
     def round_once(tick):
         fa = produce.party("alice").remote(tick)

@@ -68,6 +68,8 @@ class GpuDataPlane:
         self._own_ipc: dict = {}
         self._ipc_open_cache: dict = {}
         self._pending_lazies: list = []  # LazyIpcTensors made by decode()
+        # Persistent shared-arena slabs of this process: (lo, hi, handle).
+        self._arena_slabs: list = []
 
     def _bind_device(self):
         """hipSetDevice is per-THREAD: transport pool threads and the C++
@@ -804,6 +806,80 @@ class GpuDataPlane:
             crc = int(crc_out[2].item()) & 0xFFFFFFFF
         return seg, crc, lambda: pool.release(seg)
 
+    # -- persistent shared arena (zero-pack sends) ---------------------------
+    def alloc_shared_arena(self, nbytes: int) -> "SharedArena":
+        """Allocate ``nbytes`` of device memory in IPC-shared slabs that
+        tensors can LIVE in permanently.  A tensor backed by arena memory is
+        sent cross-party as just (handle, offset, hash) — no pack copy:
+        the receiver reads the live slab directly (materialize or fused
+        combine).  The caller must not mutate the bytes between send and
+        the peer's consume ack (the fed round structure serializes this)."""
+        self._bind_device()
+        return SharedArena(self, nbytes)
+
+    def register_arena_slab(self, ptr: int, handle: bytes, size: int) -> None:
+        self._arena_slabs.append((ptr, ptr + size, handle))
+
+    def unregister_arena_slab(self, handle: bytes) -> None:
+        self._arena_slabs = [s for s in self._arena_slabs if s[2] != handle]
+
+    def arena_lookup(self, t: "torch.Tensor"):
+        """(handle, offset) when ``t``'s bytes live wholly inside one arena
+        slab of THIS process, else None."""
+        if not self._arena_slabs or t.device.type != "cuda":
+            return None
+        p = t.data_ptr()
+        e = p + t.numel() * t.element_size()
+        for lo, hi, handle in self._arena_slabs:
+            if lo <= p and e <= hi:
+                return handle, p - lo
+        return None
+
+    def arena_checksum(self, t: "torch.Tensor"):
+        """Checksum of an arena-resident tensor (encode-time; ordered after
+        the producing stream)."""
+        self._bind_device()
+        flat = t.view(-1).view(torch.uint8)
+        kind, out = self._ck_async(flat)
+        torch.cuda.current_stream(self.device).synchronize()
+        return kind, self._ck_value(kind, out)
+
+    def materialize_region(self, man, dtype, shape) -> "torch.Tensor":
+        """Copy-out + verify of an arena region manifest ({ipcp, off,
+        nbytes, crc32?, ck?})."""
+        group = {"slabs": [bytes(man["ipcp"])]}
+        man2 = {
+            "slab": 0, "off": man["off"], "nbytes": man["nbytes"],
+            "crc32": man.get("crc32"), "ck": man.get("ck", "crc32"),
+        }
+        return self.unpack_from_ipc_group(group, man2, dtype, shape)
+
+    def combine_from_region(self, man, local_flat, out_flat, wa, wb):
+        """Fused combine+verify reading an arena region in place (the
+        zero-pack FedAvg path).  bf16, fnv64-checksummed regions only."""
+        self._bind_device()
+        nbytes = man["nbytes"]
+        off = man["off"]
+        expect = man.get("crc32")
+        if expect is not None and man.get("ck") != "fnv64":
+            raise ValueError("combine_from_region requires fnv64 checksums")
+        src = self._ipc_src_view(bytes(man["ipcp"]), off + nbytes)[
+            off : off + nbytes
+        ]
+        hash_out = self._ext.fedavg_combine_hash_async(
+            out_flat.view(-1).view(torch.uint8).view(torch.bfloat16),
+            local_flat.view(-1).view(torch.uint8).view(torch.bfloat16),
+            src, wa, wb,
+        )
+        torch.cuda.current_stream(self.device).synchronize()
+        if expect is not None and self.config.verify_crc:
+            got = int(hash_out.item()) & 0xFFFFFFFFFFFFFFFF
+            if got != expect:
+                raise ValueError(
+                    f"GPU tensor checksum mismatch (arena combine): "
+                    f"expected {expect:#x}, got {got:#x}"
+                )
+
     # -- zero-copy IPC receive (lazy_ipc) ------------------------------------
     def combine_from_ipc(self, man, local_flat: "torch.Tensor",
                          out_flat: "torch.Tensor", wa: float, wb: float):
@@ -1009,6 +1085,10 @@ class LazyIpcTensor:
     def materialize(self) -> "torch.Tensor":
         """D2D copy + verify into a regular tensor, then release."""
         try:
+            if "ipcp" in self.man:
+                return self._plane.materialize_region(
+                    self.man, self.dtype, self.shape
+                )
             return self._plane.unpack_from_ipc(self.man, self.dtype, self.shape)
         finally:
             self.release()
@@ -1016,6 +1096,11 @@ class LazyIpcTensor:
     def combine_into(self, out, local, wa: float, wb: float) -> None:
         """out = wa*local + wb*self — fused combine+verify, then release."""
         try:
+            if "ipcp" in self.man:
+                self._plane.combine_from_region(
+                    self.man, local, out.view(-1), wa, wb
+                )
+                return
             self._plane.combine_from_ipc(
                 self.man, local, out.view(-1), wa, wb
             )
@@ -1047,6 +1132,72 @@ class LazyIpcTensor:
     def __repr__(self):
         return (f"LazyIpcTensor(shape={self.shape}, dtype={self.dtype}, "
                 f"released={self._released})")
+
+
+class SharedArena:
+    """Persistent IPC-shared device memory for zero-pack cross-party sends.
+
+    Gradients (or any tensors) allocated from the arena are exchanged by
+    reference: encode ships (slab handle, offset, hash64) instead of packing
+    a staging copy — the receiver materializes or combines straight from
+    this memory.  Slabs are <= 1 GiB each (ipc_open of >= 2 GiB hangs this
+    driver) so a tensor must fit one slab; :meth:`place` lays out a list of
+    shapes shard by shard (8-byte aligned, no slab straddling).
+    """
+
+    SLAB = GpuDataPlane.IPC_SLAB_BYTES
+
+    def __init__(self, plane: GpuDataPlane, nbytes: int):
+        self._plane = plane
+        self._slabs = []  # (ptr, handle, size, uint8 view)
+        left = nbytes
+        while left > 0:
+            size = min(self.SLAB, left)
+            ptr, handle = plane._ext.ipc_alloc(size)
+            view = plane._ext.tensor_from_ptr(ptr, size, plane.device.index)
+            h = bytes(handle)
+            self._slabs.append((ptr, h, size, view))
+            plane.register_arena_slab(ptr, h, size)
+            left -= size
+
+    @property
+    def shards(self):
+        """The raw uint8 slab views."""
+        return [s[3] for s in self._slabs]
+
+    def place(self, shapes, dtype) -> list:
+        """Lay ``shapes`` out across the slabs (8-byte aligned, never
+        straddling); returns tensors viewing arena memory, in order."""
+        import math
+
+        esize = torch.empty(0, dtype=dtype).element_size()
+        out = []
+        si, off = 0, 0
+        for sh in shapes:
+            n = int(math.prod(sh)) if sh else 1
+            nb = n * esize
+            if nb > self.SLAB:
+                raise ValueError(
+                    f"tensor of {nb} bytes exceeds the {self.SLAB}-byte slab"
+                )
+            if off + nb > self._slabs[si][2]:
+                si += 1
+                off = 0
+                if si >= len(self._slabs):
+                    raise ValueError("arena too small for this placement")
+            view = self._slabs[si][3][off : off + nb]
+            out.append(view.view(dtype).view(sh))
+            off = (off + nb + 7) & ~7
+        return out
+
+    def free(self) -> None:
+        for ptr, h, _size, _v in self._slabs:
+            self._plane.unregister_arena_slab(h)
+            try:
+                self._plane._ext.ipc_free(ptr)
+            except Exception:  # noqa: BLE001
+                logger.warning("arena slab free failed", exc_info=True)
+        self._slabs = []
 
 
 _plane: Optional[GpuDataPlane] = None

@@ -177,7 +177,15 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
         dtype = _DTYPE_TO_STR.get(t.dtype)
         if dtype is None:
             raise TypeError(f"unsupported tensor dtype {t.dtype}")
-        routes.append(route_for(t, gpu_plane, shm))
+        route = route_for(t, gpu_plane, shm)
+        if (
+            route in ("ipc", "ipc_group", "shm_gpu", "shm_chunked")
+            and gpu_plane is not None
+            and gpu_plane.arena_lookup(t) is not None
+        ):
+            # Persistent shared-arena memory: send by reference, no pack.
+            route = "ipc_persist"
+        routes.append(route)
 
     group_members = [i for i, r in enumerate(routes) if r == "ipc_group"]
     if len(group_members) == 1:
@@ -205,6 +213,16 @@ def encode(obj: Any, gpu_plane=None, shm: bool = False) -> Tuple[Dict, List[memo
         }
         nbytes = man["nbytes"]
         route = routes[idx]
+        if route == "ipc_persist":
+            handle, off = gpu_plane.arena_lookup(t)
+            man["ipcp"] = handle
+            man["off"] = off
+            if gpu_plane.config.verify_crc:
+                kind, value = gpu_plane.arena_checksum(t)
+                man["crc32"] = value
+                man["ck"] = kind
+            manifests.append(man)
+            continue
         if route == "ipc_group":
             man["ipcg"] = True
             man.update(group_fields[idx])
@@ -476,6 +494,27 @@ def decode(
         nbytes = man["nbytes"]
         if man.get("ipcg"):
             tensors.append(group_batch[idx])
+            continue
+        if "ipcp" in man:
+            if gpu_plane is None:
+                raise RuntimeError(
+                    "received an arena-resident tensor but no GPU data "
+                    "plane is attached"
+                )
+            if allow_lazy and gpu_plane.config.lazy_ipc:
+                from rayfed_amd.ops.gpu_plane import LazyIpcTensor
+
+                lazy = LazyIpcTensor(
+                    gpu_plane, man, _STR_TO_DTYPE[man["dtype"]], man["shape"]
+                )
+                gpu_plane.register_lazy(lazy)
+                tensors.append(lazy)
+                continue
+            tensors.append(
+                gpu_plane.materialize_region(
+                    man, _STR_TO_DTYPE[man["dtype"]], man["shape"]
+                )
+            )
             continue
         if "ipc_slabs" in man:
             if gpu_plane is None:

@@ -365,7 +365,7 @@ class TcpReceiverProxy(base_proxy.ReceiverProxy):
                     # request body until the reader consumes it.
                     kind, header, payload = frames.decode_frame(body[8:])
                     if kind == frames.KIND_TENSOR and any(
-                        "shm" in m or "ipc_slabs" in m or m.get("ipcg")
+                        "shm" in m or "ipc_slabs" in m or "ipcp" in m or m.get("ipcg")
                         for m in header.get("tensors", ())
                     ):
                         # shm lane: consume (H2D + CRC) BEFORE acking — the

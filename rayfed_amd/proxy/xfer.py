@@ -388,7 +388,7 @@ class XferSenderService:
                 }
                 header.update(wire_header)
                 defer_ack = any(
-                    "shm" in m or "ipc_slabs" in m or m.get("ipcg")
+                    "shm" in m or "ipc_slabs" in m or "ipcp" in m or m.get("ipcg")
                     for m in extras["tensors"]
                 )
                 body_parts = [

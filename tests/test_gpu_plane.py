@@ -719,3 +719,71 @@ def test_fp8_hash64_fused_kernels(ext):
         torch.cuda.synchronize()
         assert torch.equal(back, ref)
         assert (int(hv2.item()) & 0xFFFFFFFFFFFFFFFF) == expect
+
+
+# ------------------------------------------------- persistent shared arena
+@needs_gpu
+def test_shared_arena_zero_pack_roundtrip():
+    """Tensors living in arena slabs are sent by reference ('ipcp' manifest,
+    no pack copy) and materialize identically on the receiver."""
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    arena = plane.alloc_shared_arena(64 << 20)
+    try:
+        a, b = arena.place([(1 << 20,), (513, 257)], torch.bfloat16)
+        a.uniform_(-1, 1)
+        b.uniform_(-1, 1)
+        extras, parts = tensor_codec.encode({"a": a, "b": b}, plane, shm=True)
+        mans = extras["tensors"]
+        assert all("ipcp" in m for m in mans), mans
+        assert all(m.get("ck") == "fnv64" for m in mans)
+        payload = bytes(parts[0])  # skeleton only — no tensor bytes packed
+        assert len(parts) == 1
+        out = tensor_codec.decode(extras, memoryview(payload), plane, None)
+        assert torch.equal(out["a"], a) and torch.equal(out["b"], b)
+        tensor_codec.release_parts(extras)
+    finally:
+        arena.free()
+
+
+@needs_gpu
+def test_shared_arena_lazy_combine_and_tamper():
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane, LazyIpcTensor
+    from rayfed_amd.parallel.fedavg import weighted_combine_
+
+    plane = GpuDataPlane(GpuDataPlaneConfig(lazy_ipc=True))
+    arena = plane.alloc_shared_arena(32 << 20)
+    try:
+        (peer,) = arena.place([(4 << 20,)], torch.bfloat16)
+        peer.uniform_(-1, 1)
+        local = torch.randn(4 << 20, dtype=torch.bfloat16, device="cuda")
+        extras, parts = tensor_codec.encode(peer, plane, shm=True)
+        obj = tensor_codec.decode(
+            extras, memoryview(bytes(parts[0])), plane, None, allow_lazy=True
+        )
+        assert isinstance(obj, LazyIpcTensor) and "ipcp" in obj.man
+        plane.pop_pending_lazies()
+        out = torch.empty_like(local)
+        weighted_combine_(out, [local, obj], [0.25, 0.75])
+        ref = (0.25 * local.float() + 0.75 * peer.float()).to(torch.bfloat16)
+        assert torch.equal(out, ref)
+        tensor_codec.release_parts(extras)
+
+        # Mutating the arena AFTER the hash was taken must be detected.
+        extras2, parts2 = tensor_codec.encode(peer, plane, shm=True)
+        peer.view(torch.uint8)[777] ^= 0xFF
+        obj2 = tensor_codec.decode(
+            extras2, memoryview(bytes(parts2[0])), plane, None,
+            allow_lazy=True,
+        )
+        plane.pop_pending_lazies()
+        with pytest.raises(ValueError, match="checksum mismatch"):
+            weighted_combine_(out, [local, obj2], [0.5, 0.5])
+        tensor_codec.release_parts(extras2)
+    finally:
+        arena.free()
