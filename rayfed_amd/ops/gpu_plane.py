@@ -511,6 +511,9 @@ class GpuDataPlane:
         own = self._own_ipc.get(handle)
         if own is not None:  # same-process loopback: use the local mapping
             return own[3][:nbytes]
+        for lo, _hi, h in self._arena_slabs:  # own arena slab, same process
+            if h == handle:
+                return self._ext.tensor_from_ptr(lo, nbytes, self.device.index)
         with self._lock:
             ptr = self._ipc_open_cache.get(handle)
             if ptr is None:

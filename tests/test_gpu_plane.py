@@ -733,17 +733,27 @@ def test_shared_arena_zero_pack_roundtrip():
     plane = GpuDataPlane(GpuDataPlaneConfig())
     arena = plane.alloc_shared_arena(64 << 20)
     try:
-        a, b = arena.place([(1 << 20,), (513, 257)], torch.bfloat16)
+        # Two >= 1 MiB tensors ride by reference; a tiny one stays in the
+        # pickle payload (reference-shipping a 1 KiB tensor would cost more
+        # than inlining it).
+        a, b, c = arena.place(
+            [(1 << 20,), (1025, 513), (64,)], torch.bfloat16
+        )
         a.uniform_(-1, 1)
         b.uniform_(-1, 1)
-        extras, parts = tensor_codec.encode({"a": a, "b": b}, plane, shm=True)
+        c.uniform_(-1, 1)
+        extras, parts = tensor_codec.encode(
+            {"a": a, "b": b, "c": c}, plane, shm=True
+        )
         mans = extras["tensors"]
-        assert all("ipcp" in m for m in mans), mans
-        assert all(m.get("ck") == "fnv64" for m in mans)
-        payload = bytes(parts[0])  # skeleton only — no tensor bytes packed
-        assert len(parts) == 1
+        assert sum("ipcp" in m for m in mans) == 2, mans
+        assert all(
+            m.get("ck") == "fnv64" for m in mans if "ipcp" in m
+        )
+        payload = b"".join(bytes(p) for p in parts)
         out = tensor_codec.decode(extras, memoryview(payload), plane, None)
         assert torch.equal(out["a"], a) and torch.equal(out["b"], b)
+        assert torch.equal(out["c"], c)
         tensor_codec.release_parts(extras)
     finally:
         arena.free()
