@@ -567,6 +567,11 @@ def _run_extras() -> dict:
         ("push", ["--mode", "push", "--steps", "8", "--warmup", "2"], 360, {}),
         ("fedavg", ["--mode", "fedavg", "--steps", "5", "--warmup", "1"], 360,
          {}),
+        # Zero-pack variant: gradients live in a persistent IPC-shared
+        # arena; the exchange ships (slab handle, hash) per shard.
+        ("fedavg_arena",
+         ["--mode", "fedavg", "--steps", "5", "--warmup", "1"], 360,
+         {"RAYFED_BENCH_ARENA": "1"}),
         # Cross-host-shaped lane: same push with the same-host fast lanes
         # disabled, so the striped/chunk-streamed socket path is measured.
         ("socket_push",

@@ -797,3 +797,28 @@ def test_shared_arena_lazy_combine_and_tamper():
         tensor_codec.release_parts(extras2)
     finally:
         arena.free()
+
+
+@needs_gpu
+def test_shared_arena_mutation_across_rounds():
+    """Round k+1 sends the SAME arena tensor with new contents: the fresh
+    hash rides the new manifest and the receiver sees the new values."""
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    arena = plane.alloc_shared_arena(16 << 20)
+    try:
+        (t,) = arena.place([(2 << 20,)], torch.bfloat16)
+        for r in range(3):
+            t.fill_(float(r + 1))
+            extras, parts = tensor_codec.encode(t, plane, shm=True)
+            out = tensor_codec.decode(
+                extras, memoryview(b"".join(bytes(p) for p in parts)),
+                plane, None,
+            )
+            assert torch.equal(out, t), f"round {r}"
+            tensor_codec.release_parts(extras)
+    finally:
+        arena.free()
