@@ -20,12 +20,12 @@ Usage is a drop-in for the reference::
 Reference API surface: /root/reference/fed/__init__.py:15-30.
 """
 
-from rayfed_amd.api import get, init, kill, remote, shutdown
+from rayfed_amd.api import get, init, kill, remote, shutdown, stats
 from rayfed_amd.exceptions import FedRemoteError
 from rayfed_amd.fed_object import FedObject
 from rayfed_amd.proxy.barriers import recv, send
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "get",
@@ -33,6 +33,7 @@ __all__ = [
     "kill",
     "remote",
     "shutdown",
+    "stats",
     "send",
     "recv",
     "FedObject",

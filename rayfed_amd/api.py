@@ -223,6 +223,24 @@ def init(
         barriers.ping_others(addresses=addresses, self_party=party)
 
 
+def stats() -> Dict:
+    """Cross-silo transfer statistics for this party: sender op count and
+    per-edge bytes/latency/error counters, plus the receive op count.
+    Goes beyond the reference's ``_get_stats`` op counters
+    (/root/reference/fed/proxy/barriers.py:132-154)."""
+    out: Dict = {}
+    if barriers._sender_service is not None:
+        out["send"] = barriers._sender_service._get_stats()
+    if barriers._receiver_service is not None:
+        svc = barriers._receiver_service
+        out["recv"] = (
+            svc._get_stats()
+            if hasattr(svc, "_get_stats")
+            else {"receive_op_count": getattr(svc.proxy, "received_op_count", 0)}
+        )
+    return out
+
+
 def shutdown():
     """Intended shutdown: flush pending cross-party sends, then tear down."""
     _shutdown(intended=True)

@@ -135,3 +135,29 @@ def test_examples_demo_runs():
     assert pb.returncode == 0, out_b
     assert "The result in party alice is 5" in pa.stdout
     assert "The result in party bob is 5" in out_b
+
+
+def _driver_stats(party, addresses):
+    import rayfed_amd as fed
+
+    fed.init(addresses=addresses, party=party, logging_level="warning")
+
+    @fed.remote
+    def f(x):
+        return x * 2
+
+    o = f.party("alice").remote(3)
+    r = f.party("bob").remote(o)
+    assert fed.get(r) == 12
+    s = fed.stats()
+    assert s["send"]["send_op_count"] >= 1
+    assert "edges" in s["send"]
+    assert s["recv"]["receive_op_count"] >= 1
+    fed.shutdown()
+
+
+def test_stats_accessor():
+    """fed.stats() exposes the per-edge transfer counters ([new] API)."""
+    from tests._util import run_parties
+
+    run_parties(_driver_stats)

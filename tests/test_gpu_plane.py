@@ -822,3 +822,42 @@ def test_shared_arena_mutation_across_rounds():
             tensor_codec.release_parts(extras)
     finally:
         arena.free()
+
+
+@needs_gpu
+def test_arena_over_asyncio_transport(monkeypatch):
+    """Arena (ipcp) frames over the ASYNCIO lane: the consume-before-ack
+    path materializes the region eagerly — transport matrix for the
+    zero-pack route."""
+    monkeypatch.setenv("RAYFED_TRANSPORT", "asyncio")
+    import rayfed_amd.proxy.barriers as barriers
+    from rayfed_amd._private.global_context import (
+        clear_global_context,
+        init_global_context,
+    )
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+    from tests._util import make_addresses
+
+    addrs = make_addresses(["alice"])
+    init_global_context(current_party="alice", job_name="arena_job")
+    receiver = barriers.start_receiver_proxy(
+        addrs, "alice", job_name="arena_job", proxy_config=None
+    )
+    sender = barriers.start_sender_proxy(
+        addrs, "alice", job_name="arena_job", proxy_config=None
+    )
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    sender.proxy.gpu_plane = plane
+    receiver.proxy.gpu_plane = plane
+    arena = plane.alloc_shared_arena(16 << 20)
+    try:
+        (t,) = arena.place([(2 << 20,)], torch.bfloat16)
+        t.uniform_(-1, 1)
+        assert sender.send("alice", t, "900", "900").result(timeout=60)
+        out = receiver.get_data("alice", "900", "900").result(timeout=60)
+        assert torch.equal(out, t)
+    finally:
+        arena.free()
+        clear_global_context()
+        barriers._cleanup_proxies()
