@@ -321,7 +321,13 @@ def _fedavg_driver(party, addresses, steps, warmup, device, job_name, layers,
     # instead — the exchange ships (handle, hash) per shard, zero pack.
     shapes = llama3_8b_grad_shapes(layers, vocab)
     total = sum(int(torch.prod(torch.tensor(sh))) for sh in shapes)
-    use_arena = os.environ.get("RAYFED_BENCH_ARENA") == "1" and use_gpu
+    # Arena mode is a single-node N=1 measurement variant: torchrun member
+    # ranks mirror the flat-buffer collective, so leaders must match.
+    use_arena = (
+        os.environ.get("RAYFED_BENCH_ARENA") == "1"
+        and use_gpu
+        and os.environ.get("WORLD_SIZE", "1") == "1"
+    )
     if use_arena:
         from rayfed_amd.ops.gpu_plane import maybe_create_gpu_plane
 
