@@ -427,7 +427,7 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
     import torch
     import torch.distributed as dist
 
-
+    _pin_cpus(local_rank, world)
     dist.init_process_group(backend="gloo", rank=rank, world_size=world)
     half = world // 2
     party = "alice" if rank < half else "bob"
