@@ -861,3 +861,37 @@ def test_arena_over_asyncio_transport(monkeypatch):
         arena.free()
         clear_global_context()
         barriers._cleanup_proxies()
+
+
+@needs_gpu
+def test_vram_stable_over_repeated_pushes():
+    """30 consecutive 1 GiB pushes through the IPC lane must not grow VRAM:
+    slabs, staging and lazy handles all recycle (leak canary — the round-2
+    double-encode leak would have tripped this)."""
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops import tensor_codec
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    t = torch.randn(1 << 29, dtype=torch.bfloat16, device="cuda")  # 1 GiB
+    # Warm: pools reach steady state.
+    for _ in range(3):
+        extras, parts = tensor_codec.encode(t, plane, shm=True)
+        out = tensor_codec.decode(
+            extras, memoryview(b"".join(bytes(p) for p in parts)), plane, None
+        )
+        del out
+        tensor_codec.release_parts(extras)
+    torch.cuda.synchronize()
+    free0, _total = torch.cuda.mem_get_info()
+    for _ in range(30):
+        extras, parts = tensor_codec.encode(t, plane, shm=True)
+        out = tensor_codec.decode(
+            extras, memoryview(b"".join(bytes(p) for p in parts)), plane, None
+        )
+        del out
+        tensor_codec.release_parts(extras)
+    torch.cuda.synchronize()
+    free1, _total = torch.cuda.mem_get_info()
+    grown = max(0, free0 - free1)
+    assert grown < (2 << 30), f"VRAM grew by {grown/2**30:.2f} GiB over 30 pushes"
