@@ -297,3 +297,56 @@ def test_gil_switch_interval_tuned_and_restored():
     finally:
         fed.shutdown()
     assert abs(sys.getswitchinterval() - base) < 1e-9
+
+
+def test_chunk_stream_aborted_by_error_frame():
+    """A chunk-streamed receive whose sender dies mid-stream: the error
+    object substituted on the SAME seq ids must abort the chunk wait and
+    surface to the consumer (never hang on missing chunks)."""
+    from rayfed_amd._private import serialization
+    from rayfed_amd.exceptions import FedRemoteError
+    from rayfed_amd.proxy.grpc import frames
+    from rayfed_amd.proxy.xfer import xfer_available
+
+    if not xfer_available():
+        pytest.skip("C++ transport extension not built")
+    from rayfed_amd.proxy.xfer import XferReceiverService, XferSenderService
+
+    addrs = make_addresses(["alice"])
+    recv = XferReceiverService(addrs["alice"], "alice", "j", None)
+    send = XferSenderService(addrs, "alice", "j", None)
+    try:
+        host, port = addrs["alice"].rsplit(":", 1)
+        # Main frame announcing 3 chunks of a tensor payload...
+        inner = frames.encode_frame_prefix(
+            frames.KIND_TENSOR,
+            {"job": "j", "up": "800", "down": "800", "skel": 4,
+             "tensors": []},
+        )
+        meta = {"job": "j", "up": "800", "down": "800",
+                "xk": 3, "xc": 1 << 20, "xlen": 3 << 20}
+        main = [frames.encode_frame_prefix(frames.KIND_CHUNKED, meta), inner]
+        code, _ = send._client_ctl.send(host, int(port), "800", "800",
+                                        main, False, 30.0)
+        assert code == 200
+        # ...only chunk 0 ever arrives...
+        code, _ = send._client_bulk.send(
+            host, int(port), "800\x01x0", "800", [b"x" * (1 << 20)],
+            False, 30.0, 2, True,
+        )
+        assert code == 200
+        # ...then the sender's failure path substitutes an error object.
+        err_frame = [
+            frames.encode_frame_prefix(
+                frames.KIND_ERROR, {"job": "j", "up": "800", "down": "800"}
+            ),
+            serialization.dumps(FedRemoteError("alice", None)),
+        ]
+        code, _ = send._client_ctl.send(host, int(port), "800", "800",
+                                        err_frame, False, 30.0)
+        assert code == 200
+        with pytest.raises(FedRemoteError):
+            recv.get_data("alice", "800", "800").result(timeout=60)
+    finally:
+        send.stop()
+        recv.stop()
