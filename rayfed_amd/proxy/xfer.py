@@ -533,13 +533,25 @@ class XferSenderService:
             frames.encode_frame_prefix(frames.KIND_CHUNKED, meta),
             body_parts[0],
         ]
-        code, result = self._send_with_retry(
-            self._client_ctl, host, port, up, down, main, False
-        )
-        for f in futs:
-            c, r = f.result()
-            if c != 200 and code == 200:
-                code, result = c, r
+        try:
+            code, result = self._send_with_retry(
+                self._client_ctl, host, port, up, down, main, False
+            )
+        finally:
+            # ALWAYS join the chunk senders — the caller releases the
+            # staging these threads are still reading the moment we return
+            # (even on the main-frame failure path).
+            errs = []
+            for f in futs:
+                try:
+                    c, r = f.result()
+                except BaseException as e:  # noqa: BLE001
+                    errs.append(e)
+                    continue
+                errs.append(None if c == 200 else RuntimeError(f"[{c}] {r}"))
+        for e in errs:
+            if e is not None:
+                raise e
         return code, result
 
     def _send_with_retry(self, client, host, port, up, down, parts, defer_ack,
