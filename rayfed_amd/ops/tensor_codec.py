@@ -356,24 +356,29 @@ def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
             )
         )
     skeleton = bytearray(skel_len)
-    pend = []  # (event, pinned) — released after the final sync
-    for i, data in chunks_iter:
-        lo = i * chunk_bytes
-        hi = lo + len(data)
-        mv = memoryview(data)
-        if lo < skel_len:
-            take = min(hi, skel_len) - lo
-            skeleton[lo : lo + take] = mv[:take]
-        for (slo, shi), out in zip(spans, outs):
-            if hi <= slo or lo >= shi:
-                continue
-            s = max(lo, slo)
-            e = min(hi, shi)
-            flat = out.view(-1).view(torch.uint8)
-            pend.append(
-                gpu_plane.h2d_copy(flat, s - slo, mv[s - lo : e - lo])
-            )
-    gpu_plane.finish_h2d(pend)
+    pend = []  # (event, pinned, keepalive) — released after the final sync
+    try:
+        for i, data in chunks_iter:
+            lo = i * chunk_bytes
+            hi = lo + len(data)
+            mv = memoryview(data)
+            if lo < skel_len:
+                take = min(hi, skel_len) - lo
+                skeleton[lo : lo + take] = mv[:take]
+            for (slo, shi), out in zip(spans, outs):
+                if hi <= slo or lo >= shi:
+                    continue
+                s = max(lo, slo)
+                e = min(hi, shi)
+                flat = out.view(-1).view(torch.uint8)
+                pend.append(
+                    gpu_plane.h2d_copy(flat, s - slo, mv[s - lo : e - lo])
+                )
+    finally:
+        # Even on an aborted stream (timeout, substituted error) the
+        # in-flight DMAs must complete BEFORE their source buffers are
+        # dropped with this frame.
+        gpu_plane.finish_h2d(pend)
     # Device-side verify against the sender's wire checksum.
     if gpu_plane.config.verify_crc:
         for m, out in zip(mans, outs):
