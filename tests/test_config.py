@@ -112,3 +112,18 @@ def test_channel_option_precedence_explicit_over_max_size():
     assert options["grpc.max_send_message_length"] == 77
     # The non-overridden one still reflects messages_max_size_in_bytes.
     assert options["grpc.max_receive_message_length"] == 1000
+
+
+def test_gpu_plane_config_from_dict_filters_unknown_keys():
+    from rayfed_amd.config import GpuDataPlaneConfig
+
+    cfg = GpuDataPlaneConfig.from_dict(
+        {"chunk_bytes": 123, "lazy_ipc": True, "device_checksum": "crc32",
+         "not_a_knob": 1}
+    )
+    assert cfg.chunk_bytes == 123
+    assert cfg.lazy_ipc is True
+    assert cfg.device_checksum == "crc32"
+    assert not hasattr(cfg, "not_a_knob")
+    # Defaults preserved for unspecified fields.
+    assert cfg.verify_crc is True and cfg.place_on_gpu is True
