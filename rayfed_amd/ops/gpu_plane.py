@@ -356,27 +356,27 @@ class GpuDataPlane:
         if self.config.verify_crc:
             with torch.cuda.stream(self._crc_stream):
                 self._crc_stream.wait_event(produced)
-        fnv_group = (
-            self.config.verify_crc and self.config.device_checksum == "fnv64"
-        )
         for f, (si, off) in zip(flats, placements):
             n = f.numel()
             dstv = slabs[si][3][off : off + n]
+            if self.config.verify_crc and self.config.device_checksum == "fnv64":
+                # fnv64 arena path: fused copy+hash when the source is
+                # 8-byte aligned; otherwise copy then hash the (always
+                # aligned) slab region on the same stream — either way the
+                # LDS-bound CRC32 never runs here.
+                with torch.cuda.stream(self._copy_stream):
+                    if (f.data_ptr() & 7) == 0:
+                        out = self._ext.pack_hash64_async(f, dstv)
+                    else:
+                        dstv.copy_(f)
+                        out = self._ext.hash64_async(dstv)
+                crc_outs.append(("fnv64", out))
+                continue
             with torch.cuda.stream(self._copy_stream):
                 dstv.copy_(f)
-            if self.config.verify_crc and not fnv_group:
+            if self.config.verify_crc:
                 with torch.cuda.stream(self._crc_stream):
                     crc_outs.append(self._ck_async(f))
-        slab_crc_outs = []
-        if fnv_group:
-            # ONE checksum per SLAB over its used range (ordered after the
-            # copies on the same stream) instead of one per tensor — a
-            # 291-tensor state dict launches ~16 hash kernels, not 291.
-            with torch.cuda.stream(self._copy_stream):
-                for sl, used in zip(slabs, slab_sizes):
-                    slab_crc_outs.append(
-                        self._ext.hash64_async(sl[3][:used])
-                    )
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
         done.synchronize()
@@ -385,19 +385,22 @@ class GpuDataPlane:
                 crc_done = self._crc_stream.record_event()
             crc_done.synchronize()
         group = {"slabs": [s[1] for s in slabs], "slab_bytes": S}
-        if fnv_group:
-            group["slab_used"] = list(slab_sizes)
-            group["slab_crcs"] = [
-                int(v) & 0xFFFFFFFFFFFFFFFF
-                for v in torch.cat(slab_crc_outs).cpu().tolist()
-            ] if slab_crc_outs else []
-            group["ck"] = "fnv64"
         fields = []
+        vals = None
+        if self.config.verify_crc and crc_outs and all(
+            k == "fnv64" for k, _ in crc_outs
+        ):
+            # ONE batched D2H for the whole group's checksum values.
+            vals = torch.cat([o for _, o in crc_outs]).cpu().tolist()
         for i, (si, off) in enumerate(placements):
             f = {"slab": si, "off": off}
-            if self.config.verify_crc and not fnv_group:
+            if self.config.verify_crc:
                 kind, out = crc_outs[i]
-                f["crc32"] = self._ck_value(kind, out)
+                f["crc32"] = (
+                    int(vals[i]) & 0xFFFFFFFFFFFFFFFF
+                    if vals is not None
+                    else self._ck_value(kind, out)
+                )
                 f["ck"] = kind
             fields.append(f)
 
@@ -415,16 +418,6 @@ class GpuDataPlane:
         outs = []
         ck_outs = []  # (i, kind, out_tensor)
         crc_pairs = []  # for crc32 entries on the crc stream
-        slab_verify = []  # (slab_idx, out_tensor) for slab-level checksums
-        if group.get("slab_crcs"):
-            # Slab-level verify: one hash per slab's used range, in
-            # parallel with the copy-outs (nothing writes the slabs here).
-            with torch.cuda.stream(self._crc_stream):
-                for si, (h, used) in enumerate(
-                    zip(group["slabs"], group["slab_used"])
-                ):
-                    src = self._ipc_src_view(bytes(h), used)[:used]
-                    slab_verify.append((si, self._ext.hash64_async(src)))
         for i, (man, dtype, shape) in enumerate(zip(mans, dtypes, shapes)):
             nbytes = man["nbytes"]
             handle = bytes(group["slabs"][man["slab"]])
@@ -449,22 +442,12 @@ class GpuDataPlane:
                     crc_pairs.append(i)
         with torch.cuda.stream(self._copy_stream):
             done = self._copy_stream.record_event()
-        if crc_pairs or slab_verify:
+        if crc_pairs:
             with torch.cuda.stream(self._crc_stream):
                 crc_done = self._crc_stream.record_event()
         done.synchronize()
-        if crc_pairs or slab_verify:
+        if crc_pairs:
             crc_done.synchronize()
-        if slab_verify:
-            got = torch.cat([o for _, o in slab_verify]).cpu().tolist()
-            for (si, _o), v in zip(slab_verify, got):
-                expect = group["slab_crcs"][si]
-                if (int(v) & 0xFFFFFFFFFFFFFFFF) != expect:
-                    raise ValueError(
-                        f"GPU tensor checksum mismatch (arena group, slab "
-                        f"{si}): expected {expect:#x}, got "
-                        f"{int(v) & 0xFFFFFFFFFFFFFFFF:#x}"
-                    )
         if ck_outs:
             if all(o[1] == "fnv64" for o in ck_outs):
                 # ONE batched D2H for every checksum value.
