@@ -161,3 +161,23 @@ def test_stats_accessor():
     from tests._util import run_parties
 
     run_parties(_driver_stats)
+
+
+def test_examples_secure_agg_runs():
+    """The pairwise-masked secure-aggregation example runs end to end and
+    the masks cancel exactly."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = os.path.join(repo, "examples", "secure_agg.py")
+    pb = subprocess.Popen([sys.executable, script, "bob"],
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                          text=True, cwd=repo)
+    pa = subprocess.run([sys.executable, script, "alice"],
+                        capture_output=True, text=True, timeout=120, cwd=repo)
+    out_b, _ = pb.communicate(timeout=120)
+    assert pa.returncode == 0, pa.stdout + pa.stderr
+    assert pb.returncode == 0, out_b
+    assert "masked aggregate mean" in pa.stdout
