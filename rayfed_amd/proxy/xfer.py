@@ -68,6 +68,17 @@ def _prestart_pool(pool: ThreadPoolExecutor, n: int) -> None:
         pass
 
 
+def _chunk_stripes() -> int:
+    """Stripe fanout per sidecar chunk frame (parallel connections the
+    chunk-streamed socket lane rides); RAYFED_CHUNK_STRIPES overrides."""
+    import os
+
+    try:
+        return max(1, min(16, int(os.environ.get("RAYFED_CHUNK_STRIPES", "4"))))
+    except ValueError:
+        return 4
+
+
 def _xfer_debug() -> bool:
     import os
 
@@ -523,7 +534,7 @@ class XferSenderService:
             # starve under concurrent big sends.
             futs.append(self._chunk_pool.submit(
                 self._send_with_retry, self._client_bulk, host, port,
-                f"{up}\x01x{i}", down, sub, False, 4, True,
+                f"{up}\x01x{i}", down, sub, False, _chunk_stripes(), True,
             ))
         meta = {
             "job": self._job_name, "up": up, "down": down,
