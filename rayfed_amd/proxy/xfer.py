@@ -74,9 +74,9 @@ def _chunk_stripes() -> int:
     import os
 
     try:
-        return max(1, min(16, int(os.environ.get("RAYFED_CHUNK_STRIPES", "4"))))
+        return max(1, min(16, int(os.environ.get("RAYFED_CHUNK_STRIPES", "8"))))
     except ValueError:
-        return 4
+        return 8
 
 
 def _xfer_debug() -> bool:
