@@ -20,9 +20,16 @@ of lane r-N/2.  Rank 0 prints ONE JSON line; elapsed is the MAX over ranks.
 
 Modes: ``tiny`` (headline; ``--parties 3 --tls`` for BASELINE config 5),
 ``push`` (config 3: 4 GiB bf16 tensor push alice→bob over the device-IPC
-lane; ``RAYFED_BENCH_WIRE_FP8=1`` for fp8 wire), ``fedavg`` (config 4:
-Llama-3-8B gradients, intra-party RCCL + cross-party exchange + HIP
-combine).
+lane; ``RAYFED_BENCH_WIRE_FP8=1`` for fp8 wire; ``RAYFED_SHM=0`` for the
+cross-host-shaped socket lane), ``fedavg`` (config 4: Llama-3-8B
+gradients, intra-party RCCL + cross-party exchange + HIP combine;
+``RAYFED_BENCH_ARENA=1`` for the zero-pack shared-gradient arena).
+
+The default tiny N=1 GPU run ALSO measures push / fedavg / arena-fedavg /
+socket-push in subprocesses and reports them under ``config.extras`` so a
+single driver invocation observes every data-plane headline
+(``RAYFED_BENCH_EXTRAS=0`` skips).  The JSON line always carries
+``transport`` (cpp/asyncio/grpc) and, for data modes, the ``lane``.
 """
 from __future__ import annotations
 
