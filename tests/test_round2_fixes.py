@@ -350,3 +350,19 @@ def test_chunk_stream_aborted_by_error_frame():
     finally:
         send.stop()
         recv.stop()
+
+
+def test_chunk_stripes_env_knob(monkeypatch):
+    """RAYFED_CHUNK_STRIPES controls per-chunk fanout, clamped to [1, 16]."""
+    from rayfed_amd.proxy import xfer
+
+    monkeypatch.delenv("RAYFED_CHUNK_STRIPES", raising=False)
+    assert xfer._chunk_stripes() == 8  # measured-best default (TUNING.md)
+    monkeypatch.setenv("RAYFED_CHUNK_STRIPES", "2")
+    assert xfer._chunk_stripes() == 2
+    monkeypatch.setenv("RAYFED_CHUNK_STRIPES", "0")
+    assert xfer._chunk_stripes() == 1
+    monkeypatch.setenv("RAYFED_CHUNK_STRIPES", "99")
+    assert xfer._chunk_stripes() == 16
+    monkeypatch.setenv("RAYFED_CHUNK_STRIPES", "not-a-number")
+    assert xfer._chunk_stripes() == 8
