@@ -181,3 +181,30 @@ def test_examples_secure_agg_runs():
     assert pa.returncode == 0, pa.stdout + pa.stderr
     assert pb.returncode == 0, out_b
     assert "masked aggregate mean" in pa.stdout
+
+
+def test_doctor_preflight():
+    """`python -m rayfed_amd.doctor` passes on this box (required checks)."""
+    import subprocess
+    import sys
+
+    proc = subprocess.run(
+        [sys.executable, "-m", "rayfed_amd.doctor"],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "cpp transport" in proc.stdout
+    # Required rows must all be ok; warns are fine on CPU-only boxes.
+    for line in proc.stdout.splitlines():
+        assert not line.startswith("[FAIL]"), line
+
+
+def test_doctor_run_checks_shape():
+    from rayfed_amd import doctor
+
+    rows = doctor.run_checks()
+    names = {r[0] for r in rows}
+    assert {"python", "torch", "cpp transport", "hip kernels", "gpu"} <= names
+    for name, required, ok, detail in rows:
+        assert isinstance(required, bool) and isinstance(ok, bool)
+        assert isinstance(detail, str) and detail
