@@ -142,23 +142,31 @@ class _Connection:
         total = len(prefix) + sum(len(p) for p in parts)
         head = (total + 8).to_bytes(8, "little") + req_id.to_bytes(8, "little")
         _CHUNK = chunk  # bound transport buffering for multi-GiB parts
-        async with self.write_lock:  # one frame's bytes stay contiguous
-            if total <= 65536:
-                # Small request: one write, one TCP segment.
-                self.writer.write(head + prefix + b"".join(bytes(p) for p in parts))
-            else:
-                self.writer.write(head)
-                self.writer.write(prefix)
-                for p in parts:
-                    if len(p) <= _CHUNK:
-                        self.writer.write(p)
-                    else:
-                        mv = memoryview(p)
-                        for off in range(0, len(mv), _CHUNK):
-                            self.writer.write(mv[off : off + _CHUNK])
-                            await self.writer.drain()
-            await self.writer.drain()
-        return await asyncio.wait_for(fut, timeout=timeout)
+        try:
+            async with self.write_lock:  # one frame's bytes stay contiguous
+                if total <= 65536:
+                    # Small request: one write, one TCP segment.
+                    self.writer.write(
+                        head + prefix + b"".join(bytes(p) for p in parts)
+                    )
+                else:
+                    self.writer.write(head)
+                    self.writer.write(prefix)
+                    for p in parts:
+                        if len(p) <= _CHUNK:
+                            self.writer.write(p)
+                        else:
+                            mv = memoryview(p)
+                            for off in range(0, len(mv), _CHUNK):
+                                self.writer.write(mv[off : off + _CHUNK])
+                                await self.writer.drain()
+                await self.writer.drain()
+            return await asyncio.wait_for(fut, timeout=timeout)
+        finally:
+            # A timed-out / cancelled / failed request must not leave its
+            # future parked in ``pending`` for the connection's lifetime
+            # (the ack for a retried id would otherwise resolve a dead slot).
+            self.pending.pop(req_id, None)
 
     async def close(self):
         self.alive = False

@@ -366,3 +366,30 @@ def test_chunk_stripes_env_knob(monkeypatch):
     assert xfer._chunk_stripes() == 16
     monkeypatch.setenv("RAYFED_CHUNK_STRIPES", "not-a-number")
     assert xfer._chunk_stripes() == 8
+
+
+def test_asyncio_request_timeout_cleans_pending():
+    """A timed-out request must not leave its future in conn.pending."""
+    import asyncio
+
+    from rayfed_amd.proxy.tcp.tcp_proxy import _Connection
+
+    async def scenario():
+        async def handle(reader, writer):
+            await reader.read(1 << 16)  # swallow the frame, never ack
+            await asyncio.sleep(5)
+
+        server = await asyncio.start_server(handle, "127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        conn = _Connection(reader, writer)
+        try:
+            with pytest.raises(asyncio.TimeoutError):
+                await conn.request(b"p", [b"x" * 64], timeout=0.2)
+            assert not conn.pending  # slot reclaimed despite no ack
+        finally:
+            await conn.close()
+            server.close()
+            await server.wait_closed()
+
+    asyncio.run(scenario())
