@@ -80,3 +80,34 @@ def test_plaintext_rejected_by_tls_receiver(tls_config):
         timeout=60,
         expect_codes=[0, 0],
     )
+
+
+def _driver_tls_bulk(party, addresses, tls_config):
+    """A multi-MB numpy payload rides the TLS bulk path (striped when the
+    C++ core is active), not just the inline small-frame exchange.  The
+    same-host /dev/shm lane is disabled so the bytes really cross TLS."""
+    import numpy as np
+
+    os.environ["RAYFED_SHM"] = "0"
+    fed.init(addresses=addresses, party=party, tls_config=tls_config)
+
+    @fed.remote
+    def make():
+        rng = np.random.default_rng(7)
+        return rng.standard_normal((4, 1 << 20)).astype(np.float32)  # 16 MiB
+
+    @fed.remote
+    def check(a):
+        rng = np.random.default_rng(7)
+        want = rng.standard_normal((4, 1 << 20)).astype(np.float32)
+        assert np.array_equal(a, want)
+        return float(a.sum())
+
+    x = make.party("alice").remote()
+    s = check.party("bob").remote(x)
+    assert isinstance(fed.get(s), float)
+    fed.shutdown()
+
+
+def test_bulk_tensor_over_tls(tls_config):
+    run_parties(_driver_tls_bulk, args=(tls_config,), timeout=180)
