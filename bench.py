@@ -226,7 +226,8 @@ def _detect_lane(nbytes: int, dtype, device_type: str) -> str:
 
 
 def _push_driver(party: str, addresses, steps: int, warmup: int, device: int,
-                 job_name: str, nbytes: int, barrier_cb=None, result_q=None):
+                 job_name: str, nbytes: int, tls_config=None, barrier_cb=None,
+                 result_q=None):
     """BASELINE config 3: bf16 tensor push alice→bob; end-to-end GB/s."""
     import torch
 
@@ -240,7 +241,7 @@ def _push_driver(party: str, addresses, steps: int, warmup: int, device: int,
     if os.environ.get("RAYFED_BENCH_WIRE_FP8") == "1":
         cfg["gpu_data_plane"] = {"wire_dtype": "fp8e4m3"}
     fed.init(addresses=addresses, party=party, job_name=job_name,
-             config=cfg, logging_level="warning")
+             config=cfg, tls_config=tls_config, logging_level="warning")
 
     numel = nbytes // 2
 
@@ -514,20 +515,20 @@ def _run_single_process(mode, steps, warmup, extra=None, parties=2, tls=False):
     names = ["alice", "bob", "carol", "dave"][:parties]
     addresses = make_addresses(names)
     ctx = multiprocessing.get_context("fork")
+    tls_config = None
+    if tls:
+        import sys as _sys
+
+        _sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        from tool.generate_tls_certs import generate
+
+        tls_config = generate("/tmp/rayfed_amd/bench-certs")
+        tls_config["target_name_override"] = "localhost"
     if mode == "fedavg":
         args_extra = tuple(extra)  # (layers, vocab)
     elif mode == "push":
-        args_extra = (extra,)
+        args_extra = (extra, tls_config)
     else:
-        tls_config = None
-        if tls:
-            import sys as _sys
-
-            _sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
-            from tool.generate_tls_certs import generate
-
-            tls_config = generate("/tmp/rayfed_amd/bench-certs")
-            tls_config["target_name_override"] = "localhost"
         args_extra = (tls_config,)
     procs = [
         ctx.Process(
@@ -728,7 +729,8 @@ def main():
         metric = "cross_party_tensor_push_GBps"
         unit = "GB/s"
         config = {
-            "model": f"{args.push_gib} GiB bf16 tensor push alice->bob",
+            "model": f"{args.push_gib} GiB bf16 tensor push alice->bob"
+                     + (", TLS" if args.tls else ""),
             "global_batch": args.steps,
             "seq_len": push_bytes // 2,
             "parallelism": f"fed2p-weak x{lanes} lanes",
