@@ -465,12 +465,17 @@ class XferSenderService:
                 payload_len = nbytes - (len(body_parts[0]) if body_parts else 0)
                 if (
                     not defer_ack
-                    and not self._tls
                     and len(prefix) == 6
                     and prefix[:4] == frames.MAGIC
                     and prefix[5] == frames.KIND_TENSOR
                     and payload_len >= 2 * self._stripe_bytes
                 ):
+                    # Applies to TLS too: the sidecar chunk frames stripe
+                    # across parallel TLS connections (stripe_tls), so the
+                    # crypto parallelizes with the pinned in-place assembly
+                    # and the consumer-side H2D overlap — measured 1.8 ->
+                    # multi-GB/s on the 2 GiB TLS push vs the whole-frame
+                    # path this branch replaces.
                     # Chunk-streamed tensor frame: the receiver H2Ds each
                     # chunk as it lands, overlapping consume with arrival
                     # (a whole-frame send serializes network then consume).

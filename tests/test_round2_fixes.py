@@ -393,3 +393,43 @@ def test_asyncio_request_timeout_cleans_pending():
             await server.wait_closed()
 
     asyncio.run(scenario())
+
+
+def test_chunk_streamed_send_roundtrip_tls(monkeypatch, tmp_path):
+    """The chunk-streamed path also rides mutual TLS: sidecar chunk frames
+    stripe across parallel TLS connections and assemble server-side after
+    decrypt, so big TLS frames get the same pipeline as plaintext."""
+    torch = pytest.importorskip("torch")
+    from rayfed_amd.proxy.xfer import xfer_available
+
+    if not xfer_available():
+        pytest.skip("C++ transport extension not built")
+    monkeypatch.setenv("RAYFED_SHM", "0")  # force the socket/payload route
+    import os as _os
+    import sys as _sys
+
+    _sys.path.insert(0, _os.path.join(_os.path.dirname(__file__), ".."))
+    from tool.generate_tls_certs import generate
+
+    tls = generate(str(tmp_path / "certs"))
+    tls["target_name_override"] = "localhost"
+    from rayfed_amd.proxy.xfer import XferReceiverService, XferSenderService
+
+    addrs = make_addresses(["alice"])
+    cfg = GrpcCrossSiloMessageConfig.from_dict(
+        {"messages_max_size_in_bytes": 1 << 20}
+    )
+    recv = XferReceiverService(addrs["alice"], "alice", "j", cfg,
+                               tls_config=tls)
+    send = XferSenderService(addrs, "alice", "j", cfg, tls_config=tls)
+    try:
+        t = torch.arange(3 << 20, dtype=torch.int16)  # 6 MiB > 2x 1 MiB
+        obj = {"w": t, "tag": 7}
+        assert send.send("alice", obj, "900", "900").result(timeout=60)
+        out = recv.get_data("alice", "900", "900").result(timeout=60)
+        assert out["tag"] == 7 and torch.equal(out["w"], t)
+        assert send.send("alice", 42, "901", "901").result(timeout=30)
+        assert recv.get_data("alice", "901", "901").result(timeout=30) == 42
+    finally:
+        send.stop()
+        recv.stop()
