@@ -895,3 +895,53 @@ def test_vram_stable_over_repeated_pushes():
     free1, _total = torch.cuda.mem_get_info()
     grown = max(0, free0 - free1)
     assert grown < (2 << 30), f"VRAM grew by {grown/2**30:.2f} GiB over 30 pushes"
+
+
+@needs_gpu
+def test_ipc_lane_over_tls_control_frames(tmp_path):
+    """Device-IPC tensor push with mutual TLS on the control channel: the
+    manifest frame (DEFER_ACK) crosses TLS, the tensor bytes stay on the
+    zero-copy device lane.  Covers the TLS x ipc cell of the lane matrix."""
+    import os as _os
+    import sys as _sys
+
+    _sys.path.insert(0, _os.path.join(_os.path.dirname(__file__), ".."))
+    from tool.generate_tls_certs import generate
+
+    tls = generate(str(tmp_path / "certs"))
+    tls["target_name_override"] = "localhost"
+
+    import rayfed_amd.proxy.barriers as barriers
+    from rayfed_amd._private.global_context import (
+        clear_global_context,
+        init_global_context,
+    )
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+    from tests._util import make_addresses
+
+    addrs = make_addresses(["alice"])
+    init_global_context(current_party="alice", job_name="tlsipc_job")
+    receiver = barriers.start_receiver_proxy(
+        addrs, "alice", job_name="tlsipc_job", tls_config=tls,
+        proxy_config=None,
+    )
+    sender = barriers.start_sender_proxy(
+        addrs, "alice", job_name="tlsipc_job", tls_config=tls,
+        proxy_config=None,
+    )
+    plane = GpuDataPlane(GpuDataPlaneConfig())
+    sender.proxy.gpu_plane = plane
+    receiver.proxy.gpu_plane = plane
+    try:
+        t = torch.randn(4 << 20, dtype=torch.bfloat16, device="cuda")
+        from rayfed_amd.ops.tensor_codec import route_for
+
+        # The payload must NOT fall to the socket: device lane applies.
+        assert route_for(t, plane, shm=True).startswith("ipc")
+        assert sender.send("alice", t, "910", "910").result(timeout=60)
+        out = receiver.get_data("alice", "910", "910").result(timeout=60)
+        assert torch.equal(out, t)
+    finally:
+        clear_global_context()
+        barriers._cleanup_proxies()
