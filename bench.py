@@ -26,7 +26,7 @@ gradients, intra-party RCCL + cross-party exchange + HIP combine;
 ``RAYFED_BENCH_ARENA=1`` for the zero-pack shared-gradient arena).
 
 The default tiny N=1 GPU run ALSO measures push / fedavg / arena-fedavg /
-socket-push in subprocesses and reports them under ``config.extras`` so a
+socket-push / tls-push in subprocesses and reports them under ``config.extras`` so a
 single driver invocation observes every data-plane headline
 (``RAYFED_BENCH_EXTRAS=0`` skips).  The JSON line always carries
 ``transport`` (cpp/asyncio/grpc) and, for data modes, the ``lane``.
@@ -609,6 +609,11 @@ def _run_extras() -> dict:
         ("socket_push",
          ["--mode", "push", "--steps", "4", "--warmup", "1",
           "--push-gib", "2"], 360, {"RAYFED_SHM": "0"}),
+        # Same shape under mutual TLS: the chunk-streamed path rides striped
+        # parallel TLS connections (crypto parallelizes with assembly).
+        ("tls_push",
+         ["--mode", "push", "--steps", "4", "--warmup", "1",
+          "--push-gib", "2", "--tls"], 360, {"RAYFED_SHM": "0"}),
     ]
     env = dict(os.environ)
     env["RAYFED_BENCH_EXTRAS"] = "0"
