@@ -1248,7 +1248,10 @@ class XferClient {
   void release_tls_conn(const std::string& key, std::unique_ptr<Stream> st) {
     std::lock_guard<std::mutex> lk(tls_mu_);
     auto& pool = tls_pool_[key];
-    if (pool.size() < 8) {
+    // Chunk-streamed sends fan out to chunk-workers x stripes concurrent
+    // TLS exchanges (8 x 8 at the defaults); retaining fewer connections
+    // than that forces fresh handshakes every round.
+    if (pool.size() < 64) {
       pool.push_back(std::move(st));
     } else {
       st->close_free();
