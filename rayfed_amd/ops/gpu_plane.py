@@ -970,6 +970,35 @@ class GpuDataPlane:
         torch.cuda.current_stream(self.device).synchronize()
         return int(out[2].item()) & 0xFFFFFFFF
 
+    def get_wire_staging(self, nbytes: int):
+        """Pooled device uint8 buffer for streamed wire bytes (fp8 lane):
+        chunks H2D into it as they arrive, then :meth:`unpack_wire_fp8`
+        expands to the destination dtype.  Returns (flat_u8, release)."""
+        self._bind_device()
+        buf = self._get_buf(self._dev_staging, nbytes, pin=False)
+        return buf, (lambda b=buf: self._put_buf(self._dev_staging, b))
+
+    def unpack_wire_fp8(self, wire_u8: "torch.Tensor", nbytes: int,
+                        out: "torch.Tensor",
+                        crc_expect: Optional[int]) -> None:
+        """Verify + expand fp8-e4m3 wire bytes (already on device) into
+        ``out`` (bf16).  Used by the chunk-streamed decode — the H2D of
+        later chunks overlaps the network; this runs once at the end."""
+        crc_out = None
+        with torch.cuda.stream(self._copy_stream):
+            if self.config.verify_crc and crc_expect is not None:
+                crc_out = self._ext.crc32_async(wire_u8[:nbytes])
+            self._ext.unpack_fp8_async(wire_u8[:nbytes], out.view(-1))
+            done = self._copy_stream.record_event()
+        done.synchronize()
+        if crc_out is not None:
+            crc = int(crc_out[2].item()) & 0xFFFFFFFF
+            if crc != crc_expect:
+                raise ValueError(
+                    f"tensor CRC mismatch (streamed fp8): expected "
+                    f"{crc_expect:#x}, got {crc:#x}"
+                )
+
     # -- recv path ------------------------------------------------------------
     def unpack_from_host(
         self,

@@ -317,9 +317,10 @@ def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
     yields ``(index, bytes)`` in ARRIVAL order; chunk i covers payload
     bytes [i*chunk_bytes, ...).  GPU-destined payload-route tensors H2D
     each chunk as it lands — the copy-in overlaps the network — and verify
-    their checksum on device afterwards.  Any layout this fast path does
-    not handle (shm/ipc manifests, fp8 wire, CPU destination) falls back
-    to assemble-then-decode.
+    their checksum on device afterwards.  fp8-e4m3 wire streams into a
+    device wire buffer and expands to bf16 in one fused pass at the end.
+    Any layout this fast path does not handle (shm/ipc manifests, CPU
+    destination) falls back to assemble-then-decode.
     """
     mans = header["tensors"]
     skel_len = header["skel"]
@@ -329,7 +330,8 @@ def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
         and gpu_plane.config.place_on_gpu
         and all(
             "shm" not in m and "ipc_slabs" not in m and not m.get("ipcg")
-            and not m.get("wire") and m["device"] == "cuda"
+            and m.get("wire") in (None, "fp8e4m3")
+            and m["device"] == "cuda"
             for m in mans
         )
         and mans
@@ -342,55 +344,71 @@ def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
         return decode(header, memoryview(buf), gpu_plane, allowed_list)
 
     # Payload spans per tensor: skeleton first, then raw bytes in order.
+    # fp8-wire tensors stream into a pooled device wire buffer and expand
+    # to bf16 after the last chunk; plain tensors stream straight into
+    # their final storage.
     spans = []
     off = skel_len
     outs = []
+    targets = []  # flat u8 view each chunk H2Ds into
+    wire_rel = []  # (man, wire_buf, nbytes, out, release) for fp8 tensors
     for m in mans:
         n = m["nbytes"]
         spans.append((off, off + n))
         off += n
-        outs.append(
-            torch.empty(
-                m["shape"], dtype=_STR_TO_DTYPE[m["dtype"]],
-                device=gpu_plane.device,
-            )
+        out = torch.empty(
+            m["shape"], dtype=_STR_TO_DTYPE[m["dtype"]],
+            device=gpu_plane.device,
         )
+        outs.append(out)
+        if m.get("wire") == "fp8e4m3":
+            buf, rel = gpu_plane.get_wire_staging(n)
+            targets.append(buf)
+            wire_rel.append((m, buf, n, out, rel))
+        else:
+            targets.append(out.view(-1).view(torch.uint8))
     skeleton = bytearray(skel_len)
     pend = []  # (event, pinned, keepalive) — released after the final sync
     try:
-        for i, data in chunks_iter:
-            lo = i * chunk_bytes
-            hi = lo + len(data)
-            mv = memoryview(data)
-            if lo < skel_len:
-                take = min(hi, skel_len) - lo
-                skeleton[lo : lo + take] = mv[:take]
-            for (slo, shi), out in zip(spans, outs):
-                if hi <= slo or lo >= shi:
+        try:
+            for i, data in chunks_iter:
+                lo = i * chunk_bytes
+                hi = lo + len(data)
+                mv = memoryview(data)
+                if lo < skel_len:
+                    take = min(hi, skel_len) - lo
+                    skeleton[lo : lo + take] = mv[:take]
+                for (slo, shi), flat in zip(spans, targets):
+                    if hi <= slo or lo >= shi:
+                        continue
+                    s = max(lo, slo)
+                    e = min(hi, shi)
+                    pend.append(
+                        gpu_plane.h2d_copy(flat, s - slo, mv[s - lo : e - lo])
+                    )
+        finally:
+            # Even on an aborted stream (timeout, substituted error) the
+            # in-flight DMAs must complete BEFORE their source buffers are
+            # dropped with this frame.
+            gpu_plane.finish_h2d(pend)
+        # Device-side verify against the sender's wire checksum; fp8 wire
+        # verifies its wire bytes inside the fused expand.
+        for m, buf, n, out, _rel in wire_rel:
+            gpu_plane.unpack_wire_fp8(buf, n, out, m.get("crc32"))
+        if gpu_plane.config.verify_crc:
+            for m, out in zip(mans, outs):
+                if m.get("crc32") is None or m.get("wire") == "fp8e4m3":
                     continue
-                s = max(lo, slo)
-                e = min(hi, shi)
                 flat = out.view(-1).view(torch.uint8)
-                pend.append(
-                    gpu_plane.h2d_copy(flat, s - slo, mv[s - lo : e - lo])
-                )
+                got = gpu_plane.device_crc32(flat)
+                if got != m["crc32"]:
+                    raise ValueError(
+                        f"tensor CRC mismatch (streamed): expected "
+                        f"{m['crc32']:#x}, got {got:#x}"
+                    )
     finally:
-        # Even on an aborted stream (timeout, substituted error) the
-        # in-flight DMAs must complete BEFORE their source buffers are
-        # dropped with this frame.
-        gpu_plane.finish_h2d(pend)
-    # Device-side verify against the sender's wire checksum.
-    if gpu_plane.config.verify_crc:
-        for m, out in zip(mans, outs):
-            if m.get("crc32") is None:
-                continue
-            flat = out.view(-1).view(torch.uint8)
-            got = gpu_plane.device_crc32(flat)
-            if got != m["crc32"]:
-                raise ValueError(
-                    f"tensor CRC mismatch (streamed): expected "
-                    f"{m['crc32']:#x}, got {got:#x}"
-                )
+        for _m, _buf, _n, _out, rel in wire_rel:
+            rel()
     _decode_ctx.tensors = outs
     try:
         from rayfed_amd._private import serialization

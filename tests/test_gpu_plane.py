@@ -945,3 +945,38 @@ def test_ipc_lane_over_tls_control_frames(tmp_path):
     finally:
         clear_global_context()
         barriers._cleanup_proxies()
+
+
+@needs_gpu
+def test_chunk_streamed_fp8_wire_fast_path(monkeypatch):
+    """fp8-e4m3 wire over the chunk-streamed socket lane: chunks H2D into
+    a device wire buffer in arrival order, then one fused expand to bf16
+    with the wire CRC verified on device (no host assembly)."""
+    monkeypatch.setenv("RAYFED_SHM", "0")
+    from rayfed_amd.config import GrpcCrossSiloMessageConfig
+    from rayfed_amd.proxy.xfer import XferReceiverService, XferSenderService
+    from tests._util import make_addresses
+
+    addrs = make_addresses(["alice"])
+    cfg = GrpcCrossSiloMessageConfig.from_dict(
+        {"messages_max_size_in_bytes": 4 << 20}
+    )
+    recv = XferReceiverService(addrs["alice"], "alice", "j", cfg)
+    send = XferSenderService(addrs, "alice", "j", cfg)
+    from rayfed_amd.config import GpuDataPlaneConfig
+    from rayfed_amd.ops.gpu_plane import GpuDataPlane
+
+    plane = GpuDataPlane(GpuDataPlaneConfig(wire_dtype="fp8e4m3"))
+    send.gpu_plane = plane
+    recv.gpu_plane = plane
+    try:
+        t = torch.randn(16 << 20, dtype=torch.bfloat16, device="cuda")  # 32 MiB
+        assert send.send("alice", {"w": t}, "610", "610").result(timeout=60)
+        out = recv.get_data("alice", "610", "610").result(timeout=60)
+        assert out["w"].is_cuda and out["w"].dtype == torch.bfloat16
+        # fp8 wire is lossy: compare against the fp8 round-trip of t.
+        ref = t.to(torch.float8_e4m3fn).to(torch.bfloat16)
+        assert torch.equal(out["w"], ref)
+    finally:
+        send.stop()
+        recv.stop()
