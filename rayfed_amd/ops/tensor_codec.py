@@ -361,7 +361,7 @@ def decode_streamed(header: Dict, total_len: int, chunk_bytes: int,
             device=gpu_plane.device,
         )
         outs.append(out)
-        if m.get("wire") == "fp8e4m3":
+        if m.get("wire") == "fp8e4m3" and n > 0:
             buf, rel = gpu_plane.get_wire_staging(n)
             targets.append(buf)
             wire_rel.append((m, buf, n, out, rel))
