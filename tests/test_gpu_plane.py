@@ -980,3 +980,18 @@ def test_chunk_streamed_fp8_wire_fast_path(monkeypatch):
     finally:
         send.stop()
         recv.stop()
+
+
+@needs_gpu
+def test_unpack_wire_fp8_crc_tamper(plane):
+    """The streamed fp8 expand verifies the wire CRC on device."""
+    n = 1 << 16
+    src = (torch.randn(n, device="cuda") * 2).to(torch.bfloat16)
+    wire = torch.zeros(n, dtype=torch.uint8, device="cuda")
+    out_crc = plane._ext.pack_fp8_async(src, wire)
+    torch.cuda.synchronize()
+    good = int(out_crc[2].item()) & 0xFFFFFFFF
+    out = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+    plane.unpack_wire_fp8(wire, n, out, good)  # matches: no raise
+    with pytest.raises(ValueError, match="CRC"):
+        plane.unpack_wire_fp8(wire, n, out, good ^ 0xDEAD)
