@@ -46,6 +46,7 @@
 #include <string>
 #include <thread>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace py = pybind11;
@@ -564,7 +565,14 @@ class XferServer {
     std::atomic<uint64_t> got{0};
     uint64_t total;
     uint32_t cnt;
-    Assembly(uint64_t t, uint32_t c, bool pinned) : total(t), cnt(c) {
+    // Per-stripe arrival flags: a retried frame re-sends stripes that
+    // already landed; counting them again would overshoot `got` and could
+    // deliver with a range missing (or race a writer with the delivery
+    // move).  Duplicates are drained instead.
+    std::unique_ptr<std::atomic<uint8_t>[]> seen;
+    Assembly(uint64_t t, uint32_t c, bool pinned)
+        : total(t), cnt(c), seen(new std::atomic<uint8_t>[c]) {
+      for (uint32_t i = 0; i < c; ++i) seen[i].store(0);
       if (pinned)
         pbuf = PinnedPool::inst().acquire(t);
       else
@@ -672,23 +680,48 @@ class XferServer {
           std::string key = up + '\x00' + down;
           const bool pinned = (flags & kFlagPinned) != 0;
           std::shared_ptr<Assembly> asmb;
+          bool tombstoned = false;
           {
             std::lock_guard<std::mutex> lk(asm_mu_);
             auto& slot = asm_[key];
-            if (!slot || slot->total != body_total || slot->cnt != cnt)
+            if (!slot) {
+              // A retried stripe can trail its frame's delivery; starting
+              // a fresh assembly for it would leak a half-filled buffer.
+              if (delivered_set_.count(key)) {
+                asm_.erase(key);
+                tombstoned = true;
+              } else {
+                slot = std::make_shared<Assembly>(body_total, cnt, pinned);
+              }
+            } else if (slot->total != body_total || slot->cnt != cnt) {
               slot = std::make_shared<Assembly>(body_total, cnt, pinned);
-            asmb = slot;
+            }
+            if (!tombstoned) asmb = slot;
           }
+          if (tombstoned || asmb->seen[idx].exchange(1)) {
+            // Duplicate stripe (whole-frame retry after one stripe
+            // failed) — the bytes already landed; swallow this copy.
+            if (!drain(st, payload_len)) break;
+          } else
           // Payload streams DIRECTLY into the assembly buffer at its
           // offset — stripes of one frame write disjoint ranges from
           // their own connection threads, no reassembly copy.  For pinned
           // bodies the buffer is hipHostMalloc'd, so the socket reads land
           // in DMA-able memory.
-          if (!st.read_all(asmb->base() + s_off, payload_len)) break;
-          if (asmb->got.fetch_add(payload_len) + payload_len == body_total) {
+          if (!st.read_all(asmb->base() + s_off, payload_len)) {
+            asmb->seen[idx].store(0);  // not landed: the retry must write
+            break;
+          } else if (asmb->got.fetch_add(payload_len) + payload_len ==
+                     body_total) {
             {
               std::lock_guard<std::mutex> lk(asm_mu_);
               asm_.erase(key);
+              delivered_set_.insert(key);
+              delivered_fifo_.push_back(key);
+              if (delivered_fifo_.size() > 256) {
+                delivered_set_.erase(delivered_fifo_.front());
+                delivered_fifo_.pop_front();
+              }
             }
             recv_count_.fetch_add(1);
             std::lock_guard<std::mutex> lk(mail_mu_);
@@ -774,6 +807,10 @@ class XferServer {
   std::map<std::string, std::string> mail_;
   std::mutex asm_mu_;
   std::unordered_map<std::string, std::shared_ptr<Assembly>> asm_;
+  // Recently delivered stripe keys (guarded by asm_mu_): late duplicate
+  // stripes of an already-delivered frame are drained, not reassembled.
+  std::unordered_set<std::string> delivered_set_;
+  std::deque<std::string> delivered_fifo_;
   std::unordered_map<std::string, std::shared_ptr<PinnedBuf>> pinned_mail_;
   std::atomic<uint64_t> recv_count_{0};
   std::mutex tok_mu_;

@@ -433,3 +433,68 @@ def test_chunk_streamed_send_roundtrip_tls(monkeypatch, tmp_path):
     finally:
         send.stop()
         recv.stop()
+
+
+def test_duplicate_stripes_do_not_corrupt():
+    """A retried frame re-sends stripes that already landed.  Duplicates
+    must be drained (per-stripe seen flags), never double-counted into the
+    assembly's byte total — double-counting would deliver the frame with
+    the missing stripe's range zeroed.  Late duplicates of an already
+    delivered frame are drained via the tombstone set."""
+    import socket
+    import struct
+
+    from rayfed_amd._private import serialization
+    from rayfed_amd.proxy.grpc import frames
+    from rayfed_amd.proxy.xfer import XferReceiverService, xfer_available
+
+    if not xfer_available():
+        pytest.skip("C++ transport extension not built")
+
+    addrs = make_addresses(["alice"])
+    recv = XferReceiverService(addrs["alice"], "alice", "j", None)
+    host, port = addrs["alice"].rsplit(":", 1)
+
+    obj = {"blob": b"\xab" * (1 << 20), "tag": 123}
+    body = frames.encode_frame(
+        frames.KIND_PICKLE, {"job": "j", "up": "700", "down": "700"},
+        serialization.dumps(obj),
+    )
+    half = len(body) // 2
+
+    def send_stripe(idx, lo, hi, up="700", frame=None):
+        frame = body if frame is None else frame
+        names = bytes([2, 1, len(up), len(up)]) + b"j" + up.encode() * 2
+        smeta = struct.pack("<IIQQ", idx, 2, lo, len(frame))
+        payload = frame[lo:hi]
+        total = 8 + len(names) + 24 + len(payload)
+        with socket.create_connection((host, int(port)), timeout=10) as s:
+            s.sendall(struct.pack("<QQ", total, 1) + names + smeta + payload)
+            ack = b""
+            while len(ack) < 14:
+                chunk = s.recv(14 - len(ack))
+                assert chunk, "server closed during ack"
+                ack += chunk
+            _len, _rid, code = struct.unpack("<IQH", ack)
+            assert code == 200, code
+
+    try:
+        # stripe0, DUPLICATE stripe0, stripe1 — must still deliver intact.
+        send_stripe(0, 0, half)
+        send_stripe(0, 0, half)
+        send_stripe(1, half, len(body))
+        out = recv.get_data("alice", "700", "700").result(timeout=30)
+        assert out == obj
+        # Late duplicate after delivery: drained via tombstone, server fine.
+        send_stripe(1, half, len(body))
+        # A fresh normal frame still round-trips.
+        body2 = frames.encode_frame(
+            frames.KIND_PICKLE, {"job": "j", "up": "701", "down": "701"},
+            serialization.dumps("after"),
+        )
+        h2 = len(body2) // 2
+        send_stripe(0, 0, h2, up="701", frame=body2)
+        send_stripe(1, h2, len(body2), up="701", frame=body2)
+        assert recv.get_data("alice", "701", "701").result(timeout=30) == "after"
+    finally:
+        recv.stop()
