@@ -458,6 +458,9 @@ def _run_fedavg_torchrun(steps, warmup, layers, vocab, rank, world, local_rank):
     use_gpu = _has_cuda()
     backend = "nccl" if use_gpu else "gloo"
     if use_gpu:
+        # Modulo device_count: a no-op on the 8-GPU node; lets the torchrun
+        # path run (ranks sharing GPU 0) on 1-GPU validation boxes.
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     alice_g = dist.new_group(list(range(half)), backend=backend)
     bob_g = dist.new_group(list(range(half, world)), backend=backend)
@@ -556,6 +559,11 @@ def _run_torchrun(mode, steps, warmup, push_bytes, rank, world, local_rank):
 
     _pin_cpus(local_rank, world)
     dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+    if _has_cuda():
+        import torch
+
+        # No-op on the 8-GPU node; lets 1-GPU boxes validate this path.
+        local_rank = local_rank % torch.cuda.device_count()
     lanes = world // 2
     lane = rank % lanes
     party = "alice" if rank < lanes else "bob"
