@@ -24,21 +24,27 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 _mp = multiprocessing.get_context("spawn")
 
 
-def _torchrun_bench(extra_args, world=8, timeout=420):
-    port = free_ports(1)[0]
-    env = dict(os.environ)
-    env["RAYFED_BENCH_EXTRAS"] = "0"
-    env["RAYFED_BENCH_BASE_PORT"] = str(free_ports(1)[0] + 2000)
-    cmd = [
-        sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", f"--nproc-per-node={world}",
-        "--master-addr", "127.0.0.1", "--master-port", str(port),
-        os.path.join(REPO, "bench.py"), "--gpus", str(world),
-    ] + extra_args
-    r = subprocess.run(
-        cmd, capture_output=True, text=True, timeout=timeout, env=env,
-        cwd=REPO,
-    )
+def _torchrun_bench(extra_args, world=8, timeout=420, attempts=2):
+    # Two attempts with fresh ports: the probed-free lane/master ports can
+    # be grabbed by a concurrent ephemeral connection between probe and
+    # bind (rare CI flake) — a retry re-rolls every port.
+    for _ in range(attempts):
+        port = free_ports(1)[0]
+        env = dict(os.environ)
+        env["RAYFED_BENCH_EXTRAS"] = "0"
+        env["RAYFED_BENCH_BASE_PORT"] = str(free_ports(1)[0] + 2000)
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={world}",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            os.path.join(REPO, "bench.py"), "--gpus", str(world),
+        ] + extra_args
+        r = subprocess.run(
+            cmd, capture_output=True, text=True, timeout=timeout, env=env,
+            cwd=REPO,
+        )
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, f"torchrun failed:\n{r.stdout}\n{r.stderr}"
     line = next(
         ln for ln in r.stdout.strip().splitlines() if ln.startswith("{")
